@@ -1,0 +1,220 @@
+"""CPU predict/fit parity of the six estimators against sklearn oracles on
+the reference dataset (unit tests per kernel against
+sklearn/numpy CPU oracles on the shipped pickles + CSVs)."""
+
+import os
+import pickle
+import warnings
+
+import numpy as np
+import pytest
+
+from traffic_classifier_sdn_amd.models import (
+    GaussianNB,
+    KMeans,
+    KNeighborsClassifier,
+    LogisticRegression,
+    RandomForestClassifier,
+    SVC,
+    load_model,
+)
+from traffic_classifier_sdn_amd.utils import checkpoint as ckpt
+from traffic_classifier_sdn_amd.utils.metrics import accuracy
+
+REF_MODELS = "/root/reference/models"
+needs_ref = pytest.mark.skipif(not os.path.isdir(REF_MODELS), reason="reference mount absent")
+
+
+def _sklearn_load(name):
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        return pickle.load(open(os.path.join(REF_MODELS, name), "rb"))
+
+
+# ----------------------------------------------------------------------
+# predict parity on the shipped checkpoints (sklearn 1.7 can load 4 of 6)
+# ----------------------------------------------------------------------
+
+
+@needs_ref
+def test_logistic_predict_parity(dataset):
+    X, _ = dataset
+    ours = load_model(os.path.join(REF_MODELS, "LogisticRegression"), device="cpu")
+    sk = _sklearn_load("LogisticRegression")
+    np.testing.assert_array_equal(ours.predict(X), sk.predict(X))
+
+
+@needs_ref
+def test_gnb_predict_parity(dataset):
+    X, _ = dataset
+    ours = load_model(os.path.join(REF_MODELS, "GaussianNB"), device="cpu")
+    sk = _sklearn_load("GaussianNB")
+    a = ours.predict(X)
+    b = sk.predict(X)
+    assert (a == b).mean() > 0.999
+
+
+@needs_ref
+def test_kmeans_predict_parity(dataset):
+    X, _ = dataset
+    ours = load_model(os.path.join(REF_MODELS, "KMeans_Clustering"), device="cpu")
+    sk = _sklearn_load("KMeans_Clustering")
+    sk._n_threads = 1
+    a = ours.predict(X)
+    b = sk.predict(X)
+    assert (a == b).mean() > 0.999
+
+
+@needs_ref
+def test_svc_predict_parity(dataset):
+    X, _ = dataset
+    ours = load_model(os.path.join(REF_MODELS, "SVC"), device="cpu")
+    sk = _sklearn_load("SVC")
+    a = ours.predict(X)
+    b = sk.predict(X)
+    assert (a == b).mean() > 0.999
+
+
+@needs_ref
+def test_kneighbors_predict_parity_fresh_fit(split):
+    # the shipped KNeighbors pickle is unloadable by modern sklearn; oracle
+    # is a fresh sklearn brute-force fit round-tripped through our shadow
+    # reader (same data, same k)
+    from sklearn.neighbors import KNeighborsClassifier as SK
+
+    Xtr, Xte, ytr, yte = split
+    sk = SK(n_neighbors=5, algorithm="brute").fit(Xtr, ytr.astype(str))
+    params = ckpt.params_from_shadow(ckpt.shadow_loads(pickle.dumps(sk)))
+    ours = KNeighborsClassifier.from_params(params, device="cpu")
+    a = ours.predict(Xte)
+    b = sk.predict(Xte)
+    assert (a == b).mean() > 0.999
+
+
+@needs_ref
+def test_kneighbors_shipped_checkpoint_accuracy(split):
+    # shipped checkpoint loads through the shadow reader and reproduces the
+    # published 99.30% held-out accuracy on the fit data
+    ours = load_model(os.path.join(REF_MODELS, "KNeighbors"), device="cpu")
+    Xtr, Xte, ytr, yte = split
+    acc = accuracy(yte, ours.predict(Xte))
+    # quake rows are missing from the shipped CSVs, so the 5-class slice of
+    # the test split must score in the published range
+    assert acc > 0.97
+
+
+@needs_ref
+def test_rf_predict_parity_fresh_fit(split):
+    from sklearn.ensemble import RandomForestClassifier as SK
+
+    Xtr, Xte, ytr, yte = split
+    sk = SK(n_estimators=20, random_state=0).fit(Xtr, ytr.astype(str))
+    params = ckpt.params_from_shadow(ckpt.shadow_loads(pickle.dumps(sk)))
+    ours = RandomForestClassifier.from_params(params, device="cpu")
+    a = ours.predict(Xte)
+    b = sk.predict(Xte)
+    # leaf probabilities are stored f32 (the packed GPU layout); ties in the
+    # averaged probabilities may round differently than sklearn's f64 path
+    assert (a == b).mean() > 0.999
+
+
+@needs_ref
+def test_rf_shipped_checkpoint_accuracy(split):
+    ours = load_model(os.path.join(REF_MODELS, "RandomForestClassifier"), device="cpu")
+    Xtr, Xte, ytr, yte = split
+    acc = accuracy(yte, ours.predict(Xte))
+    assert acc > 0.99  # published: 99.87%
+
+
+# ----------------------------------------------------------------------
+# fit parity (accuracy within tolerance of sklearn on the same split)
+# ----------------------------------------------------------------------
+
+
+def test_logistic_fit_accuracy(split):
+    Xtr, Xte, ytr, yte = split
+    m = LogisticRegression(device="cpu").fit(Xtr, ytr)
+    acc = accuracy(yte, m.predict(Xte))
+    assert acc > 0.95  # published: 96.47% on 6-class (5-class here)
+
+
+def test_gnb_fit_matches_sklearn(split):
+    from sklearn.naive_bayes import GaussianNB as SK
+
+    Xtr, Xte, ytr, yte = split
+    m = GaussianNB(device="cpu").fit(Xtr, ytr)
+    sk = SK().fit(Xtr, ytr.astype(str))
+    np.testing.assert_allclose(m.theta_.numpy(), sk.theta_, rtol=1e-9, atol=1e-6)
+    np.testing.assert_allclose(m.var_.numpy(), sk.var_, rtol=1e-6, atol=1e-6)
+    a = m.predict(Xte)
+    b = sk.predict(Xte)
+    assert (a == b).mean() > 0.999
+
+
+def test_kmeans_fit_inertia(dataset):
+    from sklearn.cluster import KMeans as SK
+
+    X, _ = dataset
+    m = KMeans(n_clusters=5, n_init=10, seed=0, device="cpu").fit(X)
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        sk = SK(n_clusters=5, n_init=10, random_state=0).fit(X)
+    assert m.inertia_ <= sk.inertia_ * 1.05
+    assert m.predict(X).shape == (X.shape[0],)
+
+
+def test_svc_fit_accuracy(split):
+    Xtr, Xte, ytr, yte = split
+    m = SVC(device="cpu").fit(Xtr, ytr)
+    acc = accuracy(yte, m.predict(Xte))
+    # published RBF-SVC accuracy: 85.01% on the 6-class set
+    assert acc > 0.80
+
+
+def test_svc_fit_matches_sklearn_accuracy(split):
+    from sklearn.svm import SVC as SK
+
+    Xtr, Xte, ytr, yte = split
+    m = SVC(device="cpu").fit(Xtr, ytr)
+    sk = SK().fit(Xtr, ytr.astype(str))
+    ours = accuracy(yte, m.predict(Xte))
+    theirs = accuracy(yte, sk.predict(Xte))
+    assert ours >= theirs - 0.02
+
+
+def test_rf_fit_accuracy(split):
+    Xtr, Xte, ytr, yte = split
+    m = RandomForestClassifier(n_estimators=20, seed=0, device="cpu").fit(Xtr, ytr)
+    acc = accuracy(yte, m.predict(Xte))
+    assert acc > 0.99  # published: 99.87% with 100 trees
+
+
+def test_knn_fit_accuracy(split):
+    Xtr, Xte, ytr, yte = split
+    m = KNeighborsClassifier(device="cpu").fit(Xtr, ytr)
+    acc = accuracy(yte, m.predict(Xte))
+    assert acc > 0.98  # published: 99.30%
+
+
+# ----------------------------------------------------------------------
+# misc API behaviour
+# ----------------------------------------------------------------------
+
+
+def test_predict_single_row_list(split):
+    # the reference serve loop calls predict on a nested list
+    # (traffic_classifier.py:106)
+    Xtr, Xte, ytr, yte = split
+    m = GaussianNB(device="cpu").fit(Xtr, ytr)
+    label = m.predict([list(map(float, Xte[0]))])
+    assert label.shape == (1,)
+    assert label[0] in set(np.unique(ytr))
+
+
+def test_npz_save_load_round_trip(tmp_path, split):
+    Xtr, Xte, ytr, yte = split
+    m = GaussianNB(device="cpu").fit(Xtr, ytr)
+    path = str(tmp_path / "gnb.npz")
+    m.save(path)
+    m2 = load_model(path, device="cpu")
+    np.testing.assert_array_equal(m.predict(Xte), m2.predict(Xte))

@@ -1,0 +1,95 @@
+"""k-nearest-neighbors classifier (reference estimator N3).
+
+GPU-idiomatic brute force: the KD-tree embedded in the reference checkpoint
+is not ported — a fused distance-GEMM + per-row top-k kernel replaces it
+(SURVEY.md §2.2 N3).  For reference sets too large for one GPU, the fit rows
+are sharded across ranks; predict computes per-shard top-k and merges the
+(dist, label) candidates with an all-gather (the 100M-row BASELINE config).
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, Optional
+
+import numpy as np
+import torch
+
+from .. import ops
+from ..parallel import dist
+from .base import ArrayLike, Estimator, as_tensor, encode_labels
+
+
+class KNeighborsClassifier(Estimator):
+    kind = "kneighbors"
+
+    def __init__(self, n_neighbors: int = 5, batch_rows: int = 65536, device: Optional[str] = None):
+        super().__init__(device)
+        self.n_neighbors = n_neighbors
+        self.batch_rows = batch_rows
+        self.fit_X_: Optional[torch.Tensor] = None
+        self.y_: Optional[torch.Tensor] = None
+        self.sharded_ = False
+
+    def fit(self, X: ArrayLike, y: ArrayLike, sharded: bool = False):
+        """Store the reference rows (this rank's shard when ``sharded``)."""
+        self.fit_X_ = as_tensor(X, self.device, torch.float32)
+        if sharded and dist.is_initialized():
+            classes_local = np.unique(np.asarray(y).ravel())
+            all_classes = [None] * dist.world_size()
+            torch.distributed.all_gather_object(all_classes, list(classes_local))
+            self.classes_ = np.unique(np.concatenate([np.asarray(c) for c in all_classes])).astype(object)
+            lut = {c: i for i, c in enumerate(self.classes_)}
+            y_idx = torch.tensor([lut[v] for v in np.asarray(y).ravel()], dtype=torch.int64)
+        else:
+            self.classes_, y_idx = encode_labels(y)
+        self.y_ = y_idx.to(self.device)
+        self.sharded_ = sharded and dist.is_initialized()
+        return self
+
+    def kneighbors(self, X: ArrayLike):
+        """Local-shard (dist, idx) top-k, batched over query rows."""
+        Xt = as_tensor(X, self.device, torch.float32)
+        k = min(self.n_neighbors, self.fit_X_.shape[0])
+        dists, idxs = [], []
+        for lo in range(0, Xt.shape[0], self.batch_rows):
+            d, i = ops.knn_topk(Xt[lo : lo + self.batch_rows], self.fit_X_, k)
+            dists.append(d)
+            idxs.append(i)
+        return torch.cat(dists), torch.cat(idxs)
+
+    def predict_index(self, X: ArrayLike) -> torch.Tensor:
+        Xt = as_tensor(X, self.device, torch.float32)
+        C = len(self.classes_)
+        k = self.n_neighbors
+        if not self.sharded_:
+            dist_k, idx_k = self.kneighbors(Xt)
+            return ops.knn_vote(idx_k, self.y_, C)
+        # sharded reference set: local top-k then all-gather of (dist, label)
+        # candidates, final merge on every rank
+        dist_k, idx_k = self.kneighbors(Xt)
+        lab_k = self.y_[idx_k]
+        cand_d = torch.cat(dist.allgather(dist_k), dim=1)  # (n, k*W)
+        cand_l = torch.cat(dist.allgather(lab_k), dim=1)
+        merged_d, order = torch.topk(cand_d, k, dim=1, largest=False, sorted=True)
+        merged_l = torch.gather(cand_l, 1, order)
+        counts = torch.zeros(Xt.shape[0], C, dtype=torch.int32, device=Xt.device)
+        counts.scatter_add_(1, merged_l.long(), torch.ones_like(merged_l, dtype=torch.int32))
+        return torch.argmax(counts, dim=1).to(torch.int32)
+
+    # -- checkpointing -------------------------------------------------
+    def to_params(self) -> Dict[str, Any]:
+        return {
+            "kind": self.kind,
+            "classes": np.asarray(self.classes_, dtype=object),
+            "fit_X": self.fit_X_.double().cpu().numpy(),
+            "y": self.y_.cpu().numpy(),
+            "n_neighbors": int(self.n_neighbors),
+        }
+
+    @classmethod
+    def from_params(cls, params: Dict[str, Any], device: Optional[str] = None):
+        m = cls(n_neighbors=int(params["n_neighbors"]), device=device)
+        m.classes_ = np.asarray([str(c) for c in params["classes"]], dtype=object)
+        m.fit_X_ = torch.as_tensor(np.asarray(params["fit_X"], dtype=np.float32)).to(m.device)
+        m.y_ = torch.as_tensor(np.asarray(params["y"], dtype=np.int64)).to(m.device)
+        return m

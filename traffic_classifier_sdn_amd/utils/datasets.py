@@ -1,0 +1,160 @@
+"""Dataset loading: the reference training CSVs + synthetic flow-row
+generators for benchmarking.
+
+CSV quirks handled exactly as the notebooks do: the
+game CSV is comma-delimited while the rest are tab-delimited; rows with NaN
+are dropped; the 4 cumulative counter columns are dropped for the 12-feature
+model matrix (notebooks/1_log_Kmeans.ipynb cells 8-18).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+
+from .schema import CSV_HEADER_COLUMNS, FEATURE_NAMES, LABEL_COLUMN
+
+REFERENCE_DATASET_DIR = "/root/reference/datasets"
+# compact in-repo copy (created by tools/convert_reference.py) so GPU boxes
+# and CI have the data without the read-only reference mount
+PACKED_DATASET = os.path.join(os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__)))), "data", "flows.npz")
+
+CSV_FILES = {
+    "dns": "dns_training_data.csv",
+    "game": "game_training_data.csv",
+    "ping": "ping_training_data.csv",
+    "telnet": "telnet_training_data.csv",
+    "voice": "voice_training_data.csv",
+}
+
+
+def load_csv(path: str) -> Tuple[np.ndarray, np.ndarray]:
+    """One training CSV -> (rows[n,16] float64, labels[n] object).
+
+    Delimiter is sniffed from the header line (game is comma-delimited).
+    NaN/short rows are dropped (pandas read_csv + dropna equivalent).
+    """
+    with open(path, "r") as f:
+        header = f.readline()
+        delim = "," if header.count(",") > header.count("\t") else "\t"
+        cols = header.rstrip("\n").split(delim)
+        if len(cols) != len(CSV_HEADER_COLUMNS):
+            raise ValueError(f"{path}: expected 17 columns, got {len(cols)}")
+        rows: List[List[float]] = []
+        labels: List[str] = []
+        for line in f:
+            parts = line.rstrip("\n").split(delim)
+            if len(parts) != len(cols):
+                continue
+            try:
+                vals = [float(v) for v in parts[:-1]]
+            except ValueError:
+                continue
+            if any(np.isnan(v) for v in vals):
+                continue
+            rows.append(vals)
+            labels.append(parts[-1])
+    return np.asarray(rows, dtype=np.float64), np.asarray(labels, dtype=object)
+
+
+def load_reference_dataset(
+    data_dir: Optional[str] = None, features_only: bool = True
+) -> Tuple[np.ndarray, np.ndarray]:
+    """All shipped per-class CSVs concatenated (class file order as in the
+    notebooks).  Returns (X, y); X has the 12 model features when
+    ``features_only`` (cumulative columns dropped), else all 16.
+
+    Falls back to the packed in-repo copy when the reference mount is absent
+    (e.g. on a GPU box).
+    """
+    if data_dir is None and not os.path.isdir(REFERENCE_DATASET_DIR) and os.path.exists(PACKED_DATASET):
+        return load_packed_dataset(features_only=features_only)
+    data_dir = data_dir or REFERENCE_DATASET_DIR
+    Xs, ys = [], []
+    for cls in sorted(CSV_FILES):
+        path = os.path.join(data_dir, CSV_FILES[cls])
+        if not os.path.exists(path):
+            continue
+        X, y = load_csv(path)
+        Xs.append(X)
+        ys.append(y)
+    if not Xs:
+        raise FileNotFoundError(f"no training CSVs found under {data_dir}")
+    X = np.concatenate(Xs)
+    y = np.concatenate(ys)
+    if features_only:
+        keep = [CSV_HEADER_COLUMNS.index(name) for name in FEATURE_NAMES]
+        X = X[:, keep]
+    return X, y
+
+
+def save_packed_dataset(path: str = PACKED_DATASET) -> None:
+    X, y = load_reference_dataset(features_only=False)
+    np.savez_compressed(path, X=X, y=y.astype(str))
+
+
+def load_packed_dataset(
+    path: str = PACKED_DATASET, features_only: bool = True
+) -> Tuple[np.ndarray, np.ndarray]:
+    with np.load(path, allow_pickle=False) as z:
+        X = z["X"]
+        y = np.asarray([str(v) for v in z["y"]], dtype=object)
+    if features_only:
+        keep = [CSV_HEADER_COLUMNS.index(name) for name in FEATURE_NAMES]
+        X = X[:, keep]
+    return X, y
+
+
+def train_test_split_ref(
+    X: np.ndarray, y: np.ndarray, test_size: float = 0.5, random_state: int = 101
+):
+    """The notebooks' split (sklearn train_test_split, random_state=101,
+    notebooks/2_SVM.ipynb cell 10) — reproduced via sklearn when available
+    so accuracy comparisons use the identical partition."""
+    try:
+        from sklearn.model_selection import train_test_split
+
+        return train_test_split(X, y, test_size=test_size, random_state=random_state)
+    except ImportError:  # pragma: no cover
+        rng = np.random.default_rng(random_state)
+        idx = rng.permutation(len(X))
+        cut = int(len(X) * (1 - test_size))
+        tr, te = idx[:cut], idx[cut:]
+        return X[tr], X[te], y[tr], y[te]
+
+
+def synthetic_flow_rows(
+    n: int,
+    seed: int = 0,
+    reference_X: Optional[np.ndarray] = None,
+    dtype=np.float32,
+) -> np.ndarray:
+    """Synthetic 12-feature flow-stat rows for throughput benchmarking.
+
+    When ``reference_X`` is given, rows are drawn from the empirical rows
+    with multiplicative jitter (keeps realistic tree-traversal paths and
+    feature scales); otherwise log-uniform rates in realistic ranges.
+    """
+    rng = np.random.default_rng(seed)
+    if reference_X is not None:
+        idx = rng.integers(0, reference_X.shape[0], size=n)
+        X = reference_X[idx].astype(np.float64)
+        X *= rng.uniform(0.9, 1.1, size=X.shape)
+        return X.astype(dtype)
+    # column groups: delta packets, delta bytes, pps, Bps
+    pkts = rng.poisson(30.0, size=(n, 2)).astype(np.float64)           # deltas
+    byts = pkts * rng.uniform(60, 1200, size=(n, 2))
+    pps = rng.uniform(0, 200, size=(n, 4))
+    bps = pps * rng.uniform(60, 1200, size=(n, 4))
+    X = np.empty((n, 12), dtype=np.float64)
+    X[:, 0] = pkts[:, 0]
+    X[:, 1] = byts[:, 0]
+    X[:, 2:4] = pps[:, 0:2]
+    X[:, 4:6] = bps[:, 0:2]
+    X[:, 6] = pkts[:, 1]
+    X[:, 7] = byts[:, 1]
+    X[:, 8:10] = pps[:, 2:4]
+    X[:, 10:12] = bps[:, 2:4]
+    return X.astype(dtype)

@@ -1,0 +1,172 @@
+"""HIP kernel numerics vs CPU torch oracles (ops.cpu) on identical inputs.
+All tests require an MI355X (pytest -m gpu)."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():  # collected but skipped on CPU boxes
+    pytest.skip("no GPU", allow_module_level=True)
+
+from traffic_classifier_sdn_amd.ops import cpu as oc
+from traffic_classifier_sdn_amd.ops import gpu as og
+from traffic_classifier_sdn_amd.utils.datasets import load_reference_dataset, synthetic_flow_rows
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture(scope="module")
+def X_real():
+    X, y = load_reference_dataset()
+    return torch.from_numpy(X.astype(np.float32))
+
+
+@pytest.fixture(scope="module")
+def X_syn():
+    return torch.from_numpy(synthetic_flow_rows(200_000, seed=3))
+
+
+def agree(a: torch.Tensor, b: torch.Tensor) -> float:
+    return float((a.cpu() == b.cpu()).float().mean())
+
+
+def test_gnb_kernel(X_real, X_syn):
+    C = 6
+    rng = np.random.default_rng(0)
+    theta = torch.from_numpy(rng.normal(size=(C, 12)).astype(np.float32)) * 100
+    var = torch.from_numpy(rng.uniform(0.5, 100, size=(C, 12)).astype(np.float32))
+    prior = torch.full((C,), 1.0 / C)
+    for X in (X_real, X_syn):
+        ref = oc.gnb_argmax(X, theta, var, prior)
+        got = og.gnb_argmax(X.cuda(), theta.cuda(), var.cuda(), prior.cuda())
+        assert agree(ref, got) > 0.999
+
+
+def test_linear_kernel(X_real, X_syn):
+    rng = np.random.default_rng(1)
+    W = torch.from_numpy(rng.normal(size=(6, 12)).astype(np.float32))
+    b = torch.from_numpy(rng.normal(size=(6,)).astype(np.float32))
+    for X in (X_real, X_syn):
+        ref = oc.linear_argmax(X, W, b)
+        got = og.linear_argmax(X.cuda(), W.cuda(), b.cuda())
+        assert agree(ref, got) > 0.999
+
+
+def test_kmeans_kernel(X_syn):
+    rng = np.random.default_rng(2)
+    K = 6
+    centers = torch.from_numpy(
+        synthetic_flow_rows(K, seed=5).astype(np.float32)
+    )
+    ref_l, ref_c, ref_s, ref_i = oc.kmeans_assign(X_syn, centers)
+    got_l, got_c, got_s, got_i = og.kmeans_assign(X_syn.cuda(), centers.cuda())
+    assert agree(ref_l, got_l) > 0.999
+    np.testing.assert_allclose(got_c.cpu().numpy(), ref_c.numpy(), rtol=1e-3)
+    np.testing.assert_allclose(got_s.cpu().numpy(), ref_s.numpy(), rtol=1e-3)
+    assert float(got_i) == pytest.approx(float(ref_i), rel=1e-3)
+    lab2 = og.kmeans_labels(X_syn.cuda(), centers.cuda())
+    assert agree(got_l, lab2) == 1.0
+
+
+def test_rf_kernel_shipped_forest(X_real, X_syn):
+    from traffic_classifier_sdn_amd.models import load_model
+
+    m = load_model(os.path.join(REPO, "data", "ref_models", "RandomForestClassifier.npz"), device="cpu")
+    forest = m.forest
+    for X in (X_real, X_syn):
+        ref = oc.rf_argmax(X, forest)
+        got = og.rf_argmax(X.cuda(), {k: v for k, v in forest.items()})
+        assert agree(ref, got) == 1.0
+
+
+def test_svc_kernel_shipped_model(X_real):
+    from traffic_classifier_sdn_amd.models import load_model
+
+    m = load_model(os.path.join(REPO, "data", "ref_models", "SVC.npz"), device="cpu")
+    X = X_real
+    ref = oc.svc_predict(
+        X, m.support_vectors_.float(), m.dual_coef_.float(), m.intercept_.float(),
+        m.n_support_, m.gamma_,
+    )
+    got = og.svc_predict(
+        X.cuda(), m.support_vectors_.float().cuda(), m.dual_coef_.float().cuda(),
+        m.intercept_.float().cuda(), m.n_support_.cuda(), m.gamma_,
+    )
+    assert agree(ref, got) > 0.999
+
+
+def test_knn_kernel(X_real):
+    from traffic_classifier_sdn_amd.models import load_model
+
+    m = load_model(os.path.join(REPO, "data", "ref_models", "KNeighbors.npz"), device="cpu")
+    Q = X_real[:5000]
+    R = m.fit_X_.float()
+    y = m.y_
+    ref_d, ref_i = oc.knn_topk(Q, R, 5)
+    got_d, got_i = og.knn_topk(Q.cuda(), R.cuda(), 5)
+    # expanded-form (CPU GEMM) vs direct-difference (GPU) distances differ in
+    # ulps; indices may swap only on near-ties
+    assert agree(ref_i.int(), got_i) > 0.995
+    np.testing.assert_allclose(got_d.cpu().numpy(), ref_d.numpy(), rtol=1e-3, atol=1e-2)
+    # fused vote vs CPU vote
+    _, _, got_lab = og.knn_classify(Q.cuda(), R.cuda(), y.cuda(), 5, 6)
+    ref_lab = oc.knn_vote(ref_i, y, 6)
+    assert agree(ref_lab, got_lab) > 0.999
+
+
+def test_gnb_fit_stats_kernel(X_syn):
+    y = torch.from_numpy(np.random.default_rng(4).integers(0, 6, X_syn.shape[0]))
+    Xd = X_syn.double()
+    ref = oc.gnb_fit_stats(Xd, y.long(), 6)
+    got = og.gnb_fit_stats(Xd.cuda(), y.cuda(), 6)
+    for r, g in zip(ref, got):
+        np.testing.assert_allclose(g.cpu().numpy(), r.numpy(), rtol=1e-10)
+
+
+def test_logistic_grad_kernel(X_syn):
+    rng = np.random.default_rng(5)
+    X = X_syn[:50_000].double()
+    X = (X - X.mean(0)) / (X.std(0) + 1)
+    y = torch.from_numpy(rng.integers(0, 6, X.shape[0]))
+    W = torch.from_numpy(rng.normal(size=(6, 12)))
+    b = torch.from_numpy(rng.normal(size=(6,)))
+    ref = oc.logistic_loss_grad(X, y.long(), W, b, l2=1.0)
+    got = og.logistic_loss_grad(X.cuda(), y.cuda(), W.cuda(), b.cuda(), l2=1.0)
+    assert float(got[0]) == pytest.approx(float(ref[0]), rel=1e-10)
+    np.testing.assert_allclose(got[1].cpu().numpy(), ref[1].numpy(), rtol=1e-8)
+    np.testing.assert_allclose(got[2].cpu().numpy(), ref[2].numpy(), rtol=1e-8)
+
+
+def test_flow_features_kernel():
+    rng = np.random.default_rng(6)
+    n = 10_000
+    prev = torch.from_numpy(rng.integers(0, 10_000, (n, 4)).astype(np.float64))
+    cur = prev + torch.from_numpy(rng.integers(0, 1000, (n, 4)).astype(np.float64))
+    t0 = torch.full((n, 1), 100.0, dtype=torch.float64)
+    times = torch.cat([t0 + 10, t0 + 9, t0 + 9, t0], dim=1)
+    # some rows with zero deltas (division guards)
+    times[:100, 1] = times[:100, 0]
+    ref = oc.flow_features(cur, prev, times)
+    got = og.flow_features(cur.cuda(), prev.cuda(), times.cuda())
+    np.testing.assert_allclose(got.cpu().numpy(), ref.numpy(), rtol=1e-6)
+
+
+def test_end_to_end_models_on_gpu(X_real):
+    """Every converted reference checkpoint predicts on GPU and agrees with
+    the CPU path."""
+    from traffic_classifier_sdn_amd.models import load_model
+
+    names = ["LogisticRegression", "GaussianNB", "KMeans_Clustering", "SVC",
+             "KNeighbors", "RandomForestClassifier"]
+    X = X_real.numpy()
+    for name in names:
+        path = os.path.join(REPO, "data", "ref_models", name + ".npz")
+        cpu_m = load_model(path, device="cpu")
+        gpu_m = load_model(path, device="cuda")
+        a = cpu_m.predict(X)
+        b = gpu_m.predict(X)
+        assert (a == b).mean() > 0.999, name

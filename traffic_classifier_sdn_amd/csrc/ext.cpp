@@ -1,0 +1,224 @@
+// PyTorch (ROCm) bindings for the CDNA4 kernels.  This TU is host-only;
+// every kernel lives in the .hip TUs and is reached through the extern "C"
+// launchers so the torch headers never touch device code.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include <ATen/cuda/CUDAContext.h>
+
+namespace {
+
+#define CHECK_IN(t, type)                                         \
+  TORCH_CHECK((t).is_cuda(), #t " must be a GPU tensor");         \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");     \
+  TORCH_CHECK((t).scalar_type() == type, #t " has wrong dtype");
+
+hipStream_t cur_stream() {
+  return at::cuda::getCurrentCUDAStream().stream();
+}
+
+}  // namespace
+
+extern "C" {
+void launch_gnb_predict(const float*, const float*, const float*, const float*,
+                        int*, long long, int, hipStream_t);
+void launch_linear_argmax(const float*, const float*, const float*, int*,
+                          long long, int, hipStream_t);
+void launch_kmeans_assign(const float*, const float*, int*, double*, double*,
+                          double*, long long, int, int, hipStream_t);
+void launch_rf_predict(const float*, const unsigned*, const int*, const float*,
+                       int*, long long, int, int, int, int, hipStream_t);
+void launch_svc_predict(const float*, const float*, const float*,
+                        const unsigned char*, const float*, int*, long long,
+                        int, int, float, hipStream_t);
+void launch_knn_topk(const float*, const float*, const unsigned char*, float*,
+                     int*, int*, long long, long long, int, int, long long,
+                     hipStream_t);
+void launch_gnb_fit_stats(const double*, const long long*, double*, double*,
+                          double*, long long, int, hipStream_t);
+void launch_logistic_grad(const double*, const long long*, const double*,
+                          const double*, double*, double*, long long, int,
+                          hipStream_t);
+void launch_flow_features(const double*, const double*, const double*, float*,
+                          long long, hipStream_t);
+}
+
+static torch::Tensor gnb_predict(torch::Tensor X, torch::Tensor theta,
+                                 torch::Tensor inv_var, torch::Tensor cconst) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(theta, torch::kFloat32);
+  CHECK_IN(inv_var, torch::kFloat32);
+  CHECK_IN(cconst, torch::kFloat32);
+  TORCH_CHECK(X.size(1) == 12, "X must be (n,12)");
+  const long long n = X.size(0);
+  const int C = theta.size(0);
+  auto out = torch::empty({n}, X.options().dtype(torch::kInt32));
+  launch_gnb_predict(X.data_ptr<float>(), theta.data_ptr<float>(),
+                     inv_var.data_ptr<float>(), cconst.data_ptr<float>(),
+                     out.data_ptr<int>(), n, C, cur_stream());
+  return out;
+}
+
+static torch::Tensor linear_argmax(torch::Tensor X, torch::Tensor W,
+                                   torch::Tensor b) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(W, torch::kFloat32);
+  CHECK_IN(b, torch::kFloat32);
+  TORCH_CHECK(X.size(1) == 12, "X must be (n,12)");
+  const long long n = X.size(0);
+  const int C = W.size(0);
+  auto out = torch::empty({n}, X.options().dtype(torch::kInt32));
+  launch_linear_argmax(X.data_ptr<float>(), W.data_ptr<float>(),
+                       b.data_ptr<float>(), out.data_ptr<int>(), n, C,
+                       cur_stream());
+  return out;
+}
+
+static std::vector<torch::Tensor> kmeans_assign(torch::Tensor X,
+                                                torch::Tensor centers,
+                                                bool want_update) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(centers, torch::kFloat32);
+  TORCH_CHECK(X.size(1) == 12, "X must be (n,12)");
+  const long long n = X.size(0);
+  const int K = centers.size(0);
+  auto labels = torch::empty({n}, X.options().dtype(torch::kInt32));
+  auto counts = torch::zeros({K}, X.options().dtype(torch::kFloat64));
+  auto sums = torch::zeros({K, 12}, X.options().dtype(torch::kFloat64));
+  auto inertia = torch::zeros({1}, X.options().dtype(torch::kFloat64));
+  launch_kmeans_assign(X.data_ptr<float>(), centers.data_ptr<float>(),
+                       labels.data_ptr<int>(), counts.data_ptr<double>(),
+                       sums.data_ptr<double>(), inertia.data_ptr<double>(), n,
+                       K, want_update ? 1 : 0, cur_stream());
+  return {labels, counts, sums, inertia};
+}
+
+static torch::Tensor rf_predict(torch::Tensor X, torch::Tensor nodes,
+                                torch::Tensor roots, torch::Tensor leaf_proba,
+                                int64_t n_leaves, int64_t C) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(nodes, torch::kInt32);  // packed uint2 as 2x int32
+  CHECK_IN(roots, torch::kInt32);
+  CHECK_IN(leaf_proba, torch::kFloat32);
+  TORCH_CHECK(X.size(1) == 12, "X must be (n,12)");
+  TORCH_CHECK(nodes.size(1) == 2, "nodes must be (n_nodes,2) int32");
+  const long long n = X.size(0);
+  const int n_nodes = nodes.size(0);
+  const int T = roots.size(0);
+  auto out = torch::empty({n}, X.options().dtype(torch::kInt32));
+  launch_rf_predict(X.data_ptr<float>(),
+                    reinterpret_cast<const unsigned*>(nodes.data_ptr<int>()),
+                    roots.data_ptr<int>(), leaf_proba.data_ptr<float>(),
+                    out.data_ptr<int>(), n, n_nodes, (int)n_leaves, T, (int)C,
+                    cur_stream());
+  return out;
+}
+
+static torch::Tensor svc_predict(torch::Tensor X, torch::Tensor SV,
+                                 torch::Tensor dual, torch::Tensor svclass,
+                                 torch::Tensor intercept, double gamma) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(SV, torch::kFloat32);
+  CHECK_IN(dual, torch::kFloat32);
+  CHECK_IN(svclass, torch::kUInt8);
+  CHECK_IN(intercept, torch::kFloat32);
+  TORCH_CHECK(X.size(1) == 12, "X must be (n,12)");
+  const long long n = X.size(0);
+  const int nsv = SV.size(0);
+  const int C = dual.size(0) + 1;
+  TORCH_CHECK(C >= 2 && C <= 6, "svc kernel supports 2..6 classes");
+  auto out = torch::empty({n}, X.options().dtype(torch::kInt32));
+  launch_svc_predict(X.data_ptr<float>(), SV.data_ptr<float>(),
+                     dual.data_ptr<float>(), svclass.data_ptr<unsigned char>(),
+                     intercept.data_ptr<float>(), out.data_ptr<int>(), n, nsv,
+                     C, (float)gamma, cur_stream());
+  return out;
+}
+
+static std::vector<torch::Tensor> knn_topk(torch::Tensor Q, torch::Tensor R,
+                                           c10::optional<torch::Tensor> ry,
+                                           int64_t k, int64_t C,
+                                           int64_t idx_base) {
+  CHECK_IN(Q, torch::kFloat32);
+  CHECK_IN(R, torch::kFloat32);
+  TORCH_CHECK(Q.size(1) == 12 && R.size(1) == 12, "rows must be (n,12)");
+  TORCH_CHECK(k <= 32, "k <= 32");
+  const long long nq = Q.size(0);
+  const long long nr = R.size(0);
+  auto dist = torch::empty({nq, k}, Q.options());
+  auto idx = torch::empty({nq, k}, Q.options().dtype(torch::kInt32));
+  const unsigned char* ry_ptr = nullptr;
+  torch::Tensor lab;
+  int* lab_ptr = nullptr;
+  if (ry.has_value()) {
+    CHECK_IN(ry.value(), torch::kUInt8);
+    ry_ptr = ry.value().data_ptr<unsigned char>();
+    lab = torch::empty({nq}, Q.options().dtype(torch::kInt32));
+    lab_ptr = lab.data_ptr<int>();
+  }
+  launch_knn_topk(Q.data_ptr<float>(), R.data_ptr<float>(), ry_ptr,
+                  dist.data_ptr<float>(), idx.data_ptr<int>(), lab_ptr, nq, nr,
+                  (int)k, (int)C, idx_base, cur_stream());
+  if (ry.has_value()) return {dist, idx, lab};
+  return {dist, idx};
+}
+
+static std::vector<torch::Tensor> gnb_fit_stats(torch::Tensor X,
+                                                torch::Tensor y, int64_t C) {
+  CHECK_IN(X, torch::kFloat64);
+  CHECK_IN(y, torch::kInt64);
+  TORCH_CHECK(X.size(1) == 12, "X must be (n,12)");
+  const long long n = X.size(0);
+  auto count = torch::zeros({C}, X.options());
+  auto sum = torch::zeros({C, 12}, X.options());
+  auto sumsq = torch::zeros({C, 12}, X.options());
+  launch_gnb_fit_stats(X.data_ptr<double>(), y.data_ptr<long long>(),
+                       count.data_ptr<double>(), sum.data_ptr<double>(),
+                       sumsq.data_ptr<double>(), n, (int)C, cur_stream());
+  return {count, sum, sumsq};
+}
+
+static std::vector<torch::Tensor> logistic_grad(torch::Tensor X,
+                                                torch::Tensor y,
+                                                torch::Tensor W,
+                                                torch::Tensor b) {
+  CHECK_IN(X, torch::kFloat64);
+  CHECK_IN(y, torch::kInt64);
+  CHECK_IN(W, torch::kFloat64);
+  CHECK_IN(b, torch::kFloat64);
+  const long long n = X.size(0);
+  const int C = W.size(0);
+  auto grad = torch::zeros({C, 13}, X.options());
+  auto loss = torch::zeros({1}, X.options());
+  launch_logistic_grad(X.data_ptr<double>(), y.data_ptr<long long>(),
+                       W.data_ptr<double>(), b.data_ptr<double>(),
+                       grad.data_ptr<double>(), loss.data_ptr<double>(), n, C,
+                       cur_stream());
+  return {grad, loss};
+}
+
+static torch::Tensor flow_features(torch::Tensor cur, torch::Tensor prev,
+                                   torch::Tensor times) {
+  CHECK_IN(cur, torch::kFloat64);
+  CHECK_IN(prev, torch::kFloat64);
+  CHECK_IN(times, torch::kFloat64);
+  const long long n = cur.size(0);
+  auto out = torch::empty({n, 12}, cur.options().dtype(torch::kFloat32));
+  launch_flow_features(cur.data_ptr<double>(), prev.data_ptr<double>(),
+                       times.data_ptr<double>(), out.data_ptr<float>(), n,
+                       cur_stream());
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gnb_predict", &gnb_predict, "fused GaussianNB loglik+argmax");
+  m.def("linear_argmax", &linear_argmax, "logits+argmax");
+  m.def("kmeans_assign", &kmeans_assign, "Lloyd assign + partial update");
+  m.def("rf_predict", &rf_predict, "packed-forest traversal + vote");
+  m.def("svc_predict", &svc_predict, "RBF Gram + OVO vote");
+  m.def("knn_topk", &knn_topk, "brute-force top-k (+fused vote)");
+  m.def("gnb_fit_stats", &gnb_fit_stats, "per-class sufficient stats");
+  m.def("logistic_grad", &logistic_grad, "fused CE loss+grad");
+  m.def("flow_features", &flow_features, "counters -> 12 features");
+}

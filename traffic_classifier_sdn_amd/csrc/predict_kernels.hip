@@ -1,0 +1,507 @@
+// CDNA4 predict kernels for the six estimators.
+// Hand-written for gfx950: wave64 blocks, params staged in LDS, feature rows
+// as float4 vector loads, grid-stride over rows.  All label outputs use
+// first-maximum tie-breaking to match the sklearn semantics of the CPU
+// oracles (ops/cpu.py).
+//
+// Register-pressure rule observed throughout: any per-lane array indexed by
+// a RUNTIME value would be spilled to scratch by hipcc, so every such array
+// (RF class accumulators, SVC OVO accumulators, KNN k-best lists) is sized
+// by a template parameter and only indexed inside fully-unrolled loops; the
+// per-row feature vector, whose index IS runtime (tree node feature ids),
+// lives in LDS instead.
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// Gaussian NB: fused joint-log-likelihood + argmax (reference N5).
+//   score[c] = const[c] - 0.5 * sum_j (x_j - theta[c,j])^2 * inv_var[c,j]
+// Params (C*F*2 + C floats) are broadcast via LDS.
+// ---------------------------------------------------------------------------
+__global__ void gnb_predict_kernel(const float* __restrict__ X,
+                                   const float* __restrict__ theta,
+                                   const float* __restrict__ inv_var,
+                                   const float* __restrict__ cconst,
+                                   int* __restrict__ out,
+                                   long long n, int C) {
+  constexpr int F = 12;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* s_theta = reinterpret_cast<float*>(smem);
+  float* s_ivar = s_theta + C * F;
+  float* s_const = s_ivar + C * F;
+  for (int i = threadIdx.x; i < C * F; i += blockDim.x) {
+    s_theta[i] = theta[i];
+    s_ivar[i] = inv_var[i];
+  }
+  for (int i = threadIdx.x; i < C; i += blockDim.x) s_const[i] = cconst[i];
+  __syncthreads();
+
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += stride) {
+    Row12 x = load_row12(X, row);
+    float best = -INFINITY;
+    int bi = 0;
+    for (int c = 0; c < C; ++c) {
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < F; ++j) {
+        float d = x.v[j] - s_theta[c * F + j];
+        s = fmaf(d * d, s_ivar[c * F + j], s);
+      }
+      s = s_const[c] - 0.5f * s;
+      if (s > best) { best = s; bi = c; }
+    }
+    out[row] = bi;
+  }
+}
+
+extern "C" void launch_gnb_predict(const float* X, const float* theta,
+                                   const float* inv_var, const float* cconst,
+                                   int* out, long long n, int C,
+                                   hipStream_t stream) {
+  const int block = 256;
+  size_t lds = (size_t)(2 * C * 12 + C) * sizeof(float);
+  hipLaunchKernelGGL(gnb_predict_kernel, dim3(ts_grid(n, block)), dim3(block),
+                     lds, stream, X, theta, inv_var, cconst, out, n, C);
+}
+
+// ---------------------------------------------------------------------------
+// Linear (logistic) argmax: out = argmax_c (W[c,:] . x + b[c])   (N1 serve)
+// ---------------------------------------------------------------------------
+__global__ void linear_argmax_kernel(const float* __restrict__ X,
+                                     const float* __restrict__ W,
+                                     const float* __restrict__ b,
+                                     int* __restrict__ out,
+                                     long long n, int C) {
+  constexpr int F = 12;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* s_w = reinterpret_cast<float*>(smem);
+  float* s_b = s_w + C * F;
+  for (int i = threadIdx.x; i < C * F; i += blockDim.x) s_w[i] = W[i];
+  for (int i = threadIdx.x; i < C; i += blockDim.x) s_b[i] = b[i];
+  __syncthreads();
+
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += stride) {
+    Row12 x = load_row12(X, row);
+    float best = -INFINITY;
+    int bi = 0;
+    for (int c = 0; c < C; ++c) {
+      float s = s_b[c];
+#pragma unroll
+      for (int j = 0; j < F; ++j) s = fmaf(x.v[j], s_w[c * F + j], s);
+      if (s > best) { best = s; bi = c; }
+    }
+    out[row] = bi;
+  }
+}
+
+extern "C" void launch_linear_argmax(const float* X, const float* W,
+                                     const float* b, int* out, long long n,
+                                     int C, hipStream_t stream) {
+  const int block = 256;
+  size_t lds = (size_t)(C * 12 + C) * sizeof(float);
+  hipLaunchKernelGGL(linear_argmax_kernel, dim3(ts_grid(n, block)), dim3(block),
+                     lds, stream, X, W, b, out, n, C);
+}
+
+// ---------------------------------------------------------------------------
+// KMeans assignment + partial update (N6): label = argmin_k ||x - c_k||^2,
+// with per-block LDS partial (count, sum) per cluster flushed by f64 atomics
+// (one atomic set per block, guide G12), plus inertia.
+// ---------------------------------------------------------------------------
+__global__ void kmeans_assign_kernel(const float* __restrict__ X,
+                                     const float* __restrict__ centers,
+                                     int* __restrict__ labels,
+                                     double* __restrict__ counts,   // [K]
+                                     double* __restrict__ sums,     // [K,F]
+                                     double* __restrict__ inertia,  // [1]
+                                     long long n, int K, int want_update) {
+  constexpr int F = 12;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double* s_sum = reinterpret_cast<double*>(smem);        // [K,F]
+  double* s_cnt = s_sum + K * F;                          // [K]
+  float* s_c = reinterpret_cast<float*>(s_cnt + K);       // [K,F]
+  for (int i = threadIdx.x; i < K * F; i += blockDim.x) {
+    s_c[i] = centers[i];
+    if (want_update) s_sum[i] = 0.0;
+  }
+  if (want_update)
+    for (int i = threadIdx.x; i < K; i += blockDim.x) s_cnt[i] = 0.0;
+  __syncthreads();
+
+  double local_inertia = 0.0;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += stride) {
+    Row12 x = load_row12(X, row);
+    float best = INFINITY;
+    int bi = 0;
+    for (int k = 0; k < K; ++k) {
+      float d = 0.f;
+#pragma unroll
+      for (int j = 0; j < F; ++j) {
+        float t = x.v[j] - s_c[k * F + j];
+        d = fmaf(t, t, d);
+      }
+      if (d < best) { best = d; bi = k; }
+    }
+    labels[row] = bi;
+    local_inertia += (double)best;
+    if (want_update) {
+      // LDS atomics: contention limited to the block's 256 lanes
+      atomicAdd(&s_cnt[bi], 1.0);
+#pragma unroll
+      for (int j = 0; j < F; ++j) atomicAdd(&s_sum[bi * F + j], (double)x.v[j]);
+    }
+  }
+  // inertia: wave-reduce then one atomic per wave
+  local_inertia = wave_sum(local_inertia);
+  if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(inertia, local_inertia);
+  if (want_update) {
+    __syncthreads();
+    for (int i = threadIdx.x; i < K * F; i += blockDim.x)
+      atomicAdd(&sums[i], s_sum[i]);
+    for (int i = threadIdx.x; i < K; i += blockDim.x)
+      atomicAdd(&counts[i], s_cnt[i]);
+  }
+}
+
+extern "C" void launch_kmeans_assign(const float* X, const float* centers,
+                                     int* labels, double* counts, double* sums,
+                                     double* inertia, long long n, int K,
+                                     int want_update, hipStream_t stream) {
+  const int block = 256;
+  size_t lds = (size_t)(K * 12 + K) * sizeof(double) +
+               (size_t)(K * 12) * sizeof(float);
+  hipLaunchKernelGGL(kmeans_assign_kernel, dim3(ts_grid(n, block)), dim3(block),
+                     lds, stream, X, centers, labels, counts, sums, inertia, n,
+                     K, want_update);
+}
+
+// ---------------------------------------------------------------------------
+// Random forest traversal + vote (N4) — the flagship predict kernel.
+// Packed node = uint2 { x: f32 threshold bits | leaf-prob row index,
+//                       y: (right_child_global << 8) | feature (0xff = leaf) }
+// The whole packed forest (and optionally the leaf-probability table) is
+// staged in LDS; each lane walks all trees for its row, accumulating the
+// leaf class distributions, then writes argmax.  The per-lane feature row
+// lives in LDS (node feature ids are runtime indices).
+// ---------------------------------------------------------------------------
+template <int C>
+__global__ void rf_predict_kernel(const float* __restrict__ X,
+                                  const uint2* __restrict__ nodes,
+                                  const int* __restrict__ roots,  // [T]
+                                  const float* __restrict__ leaf_proba,
+                                  int* __restrict__ out,
+                                  long long n, int n_nodes, int n_leaves,
+                                  int T, int lds_nodes, int lds_probs) {
+  constexpr int F = 12;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: [row features 256*F f32][nodes uint2][leaf probs f32]
+  float* s_x = reinterpret_cast<float*>(smem);  // [block][F]
+  uint2* s_nodes = reinterpret_cast<uint2*>(s_x + blockDim.x * F);
+  float* s_probs = reinterpret_cast<float*>(s_nodes + (lds_nodes ? n_nodes : 0));
+
+  if (lds_nodes)
+    for (int i = threadIdx.x; i < n_nodes; i += blockDim.x) s_nodes[i] = nodes[i];
+  if (lds_probs)
+    for (int i = threadIdx.x; i < n_leaves * C; i += blockDim.x)
+      s_probs[i] = leaf_proba[i];
+  __syncthreads();
+
+  const uint2* nd = lds_nodes ? s_nodes : nodes;
+  const float* lp = lds_probs ? s_probs : leaf_proba;
+  float* my_x = s_x + threadIdx.x * F;
+
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += stride) {
+    Row12 x = load_row12(X, row);
+#pragma unroll
+    for (int j = 0; j < F; ++j) my_x[j] = x.v[j];
+    float acc[C];
+#pragma unroll
+    for (int c = 0; c < C; ++c) acc[c] = 0.f;
+    for (int t = 0; t < T; ++t) {
+      int idx = roots[t];
+      while (true) {
+        uint2 node = nd[idx];
+        unsigned feat = node.y & 0xffu;
+        if (feat == 0xffu) {
+          const float* p = lp + (long long)node.x * C;
+#pragma unroll
+          for (int c = 0; c < C; ++c) acc[c] += p[c];
+          break;
+        }
+        float thr = __uint_as_float(node.x);
+        idx = (my_x[feat] <= thr) ? idx + 1 : (int)(node.y >> 8);
+      }
+    }
+    float best = -INFINITY;
+    int bi = 0;
+#pragma unroll
+    for (int c = 0; c < C; ++c)
+      if (acc[c] > best) { best = acc[c]; bi = c; }
+    out[row] = bi;
+  }
+}
+
+extern "C" void launch_rf_predict(const float* X, const unsigned* nodes,
+                                  const int* roots, const float* leaf_proba,
+                                  int* out, long long n, int n_nodes,
+                                  int n_leaves, int T, int C,
+                                  hipStream_t stream) {
+  const int block = 256;
+  // LDS budget: keep under 128 KiB so the row-feature slab plus forest fit;
+  // fall back to global (L2-resident) tables for big forests.
+  size_t row_bytes = (size_t)block * 12 * sizeof(float);
+  size_t node_bytes = (size_t)n_nodes * sizeof(uint2);
+  size_t prob_bytes = (size_t)n_leaves * C * sizeof(float);
+  size_t budget = 128 * 1024;
+  int lds_nodes = (row_bytes + node_bytes) <= budget;
+  int lds_probs = lds_nodes && (row_bytes + node_bytes + prob_bytes) <= budget;
+  size_t lds = row_bytes + (lds_nodes ? node_bytes : 0) + (lds_probs ? prob_bytes : 0);
+  dim3 grid(ts_grid(n, block));
+#define RF_CASE(CV)                                                          \
+  case CV:                                                                   \
+    hipLaunchKernelGGL((rf_predict_kernel<CV>), grid, dim3(block), lds,      \
+                       stream, X, reinterpret_cast<const uint2*>(nodes),     \
+                       roots, leaf_proba, out, n, n_nodes, n_leaves, T,      \
+                       lds_nodes, lds_probs);                                \
+    return;
+  switch (C) {
+    RF_CASE(2) RF_CASE(3) RF_CASE(4) RF_CASE(5) RF_CASE(6) RF_CASE(7)
+    RF_CASE(8) RF_CASE(12) RF_CASE(16)
+    default: break;
+  }
+#undef RF_CASE
+}
+
+// ---------------------------------------------------------------------------
+// SVC RBF + one-vs-one vote (N2).  SV tiles (features, libsvm dual rows,
+// class ids) stream through LDS; each lane keeps C*(C-1) accumulators
+//   acc[c][r] = sum over SVs of class c of dual[r][sv] * exp(-gamma*d2(x,sv))
+// then dec(i<j) = acc[i][j-1] + acc[j][i] + b[pair], majority vote.
+// The per-SV class branch is wave-UNIFORM (the sv index is uniform across
+// the block), so only the matching unrolled branch executes and all
+// accumulator indices stay compile-time.
+// ---------------------------------------------------------------------------
+#define SVC_TILE 128
+
+template <int C>
+__global__ void svc_predict_kernel(const float* __restrict__ X,
+                                   const float* __restrict__ SV,    // [nsv,F]
+                                   const float* __restrict__ dual,  // [C-1,nsv]
+                                   const unsigned char* __restrict__ svclass,
+                                   const float* __restrict__ intercept,
+                                   int* __restrict__ out,
+                                   long long n, int nsv, float gamma) {
+  constexpr int F = 12;
+  constexpr int CR = C - 1;
+  constexpr int NPAIR = C * (C - 1) / 2;
+  __shared__ __attribute__((aligned(16))) float s_sv[SVC_TILE * F];
+  __shared__ float s_dual[CR * SVC_TILE];
+  __shared__ unsigned char s_cls[SVC_TILE];
+  __shared__ float s_b[NPAIR];
+
+  for (int i = threadIdx.x; i < NPAIR; i += blockDim.x) s_b[i] = intercept[i];
+
+  long long stride = (long long)gridDim.x * blockDim.x;
+  // lock-step row tiles so the SV staging loop stays uniform per block
+  for (long long base = (long long)blockIdx.x * blockDim.x; base < n;
+       base += stride) {
+    long long row = base + threadIdx.x;
+    Row12 x;
+    if (row < n) x = load_row12(X, row);
+    float acc[C * CR];
+#pragma unroll
+    for (int i = 0; i < C * CR; ++i) acc[i] = 0.f;
+
+    for (int tile = 0; tile < nsv; tile += SVC_TILE) {
+      int cnt = min(SVC_TILE, nsv - tile);
+      __syncthreads();
+      for (int i = threadIdx.x; i < cnt * F; i += blockDim.x)
+        s_sv[i] = SV[(long long)tile * F + i];
+#pragma unroll
+      for (int r = 0; r < CR; ++r)
+        for (int i = threadIdx.x; i < cnt; i += blockDim.x)
+          s_dual[r * SVC_TILE + i] = dual[(long long)r * nsv + tile + i];
+      for (int i = threadIdx.x; i < cnt; i += blockDim.x)
+        s_cls[i] = svclass[tile + i];
+      __syncthreads();
+      if (row < n) {
+        for (int s = 0; s < cnt; ++s) {
+          float d = 0.f;
+#pragma unroll
+          for (int j = 0; j < F; ++j) {
+            float t = x.v[j] - s_sv[s * F + j];
+            d = fmaf(t, t, d);
+          }
+          float kv = __expf(-gamma * d);
+          int c = s_cls[s];  // wave-uniform
+#pragma unroll
+          for (int cc = 0; cc < C; ++cc) {
+            if (c == cc) {
+#pragma unroll
+              for (int r = 0; r < CR; ++r)
+                acc[cc * CR + r] =
+                    fmaf(s_dual[r * SVC_TILE + s], kv, acc[cc * CR + r]);
+            }
+          }
+        }
+      }
+    }
+    if (row >= n) continue;
+    // OVO vote (first max wins ties, libsvm semantics)
+    int votes[C];
+#pragma unroll
+    for (int c = 0; c < C; ++c) votes[c] = 0;
+    int p = 0;
+#pragma unroll
+    for (int i = 0; i < C; ++i)
+#pragma unroll
+      for (int j = i + 1; j < C; ++j, ++p) {
+        float dec = acc[i * CR + (j - 1)] + acc[j * CR + i] + s_b[p];
+        if (dec > 0.f) votes[i]++; else votes[j]++;
+      }
+    int best = -1, bi = 0;
+#pragma unroll
+    for (int c = 0; c < C; ++c)
+      if (votes[c] > best) { best = votes[c]; bi = c; }
+    out[row] = bi;
+  }
+}
+
+extern "C" void launch_svc_predict(const float* X, const float* SV,
+                                   const float* dual,
+                                   const unsigned char* svclass,
+                                   const float* intercept, int* out,
+                                   long long n, int nsv, int C, float gamma,
+                                   hipStream_t stream) {
+  const int block = 256;
+  dim3 grid(ts_grid(n, block));
+#define SVC_CASE(CV)                                                        \
+  case CV:                                                                  \
+    hipLaunchKernelGGL((svc_predict_kernel<CV>), grid, dim3(block), 0,      \
+                       stream, X, SV, dual, svclass, intercept, out, n,     \
+                       nsv, gamma);                                         \
+    return;
+  switch (C) {
+    SVC_CASE(2) SVC_CASE(3) SVC_CASE(4) SVC_CASE(5) SVC_CASE(6)
+    default: break;
+  }
+#undef SVC_CASE
+}
+
+// ---------------------------------------------------------------------------
+// KNN brute-force top-k + fused vote (N3).  One lane per query; reference
+// rows stream through LDS in tiles shared by the block; each lane keeps a
+// sorted k-best (dist, idx) list in registers (K compile-time; ties keep the
+// lower reference index, matching sklearn ordering).  idx_base offsets
+// emitted indices for sharded reference sets (all-gather merge keys stay
+// global).
+// ---------------------------------------------------------------------------
+#define KNN_TILE 256
+#define KNN_MAXC 8
+
+template <int K>
+__global__ void knn_topk_kernel(const float* __restrict__ Q,
+                                const float* __restrict__ R,
+                                const unsigned char* __restrict__ ry,  // may be null
+                                float* __restrict__ out_d,  // [nq,K]
+                                int* __restrict__ out_i,    // [nq,K]
+                                int* __restrict__ out_lab,  // [nq] or null
+                                long long nq, long long nr, int C,
+                                long long idx_base) {
+  constexpr int F = 12;
+  __shared__ __attribute__((aligned(16))) float s_r[KNN_TILE * F];
+  __shared__ unsigned char s_y[KNN_TILE];
+
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long base = (long long)blockIdx.x * blockDim.x; base < nq;
+       base += stride) {
+    long long q = base + threadIdx.x;
+    Row12 x;
+    if (q < nq) x = load_row12(Q, q);
+    float bd[K];
+    int bi_[K];
+    int bl[K];
+#pragma unroll
+    for (int k = 0; k < K; ++k) { bd[k] = INFINITY; bi_[k] = -1; bl[k] = 0; }
+
+    for (long long tile = 0; tile < nr; tile += KNN_TILE) {
+      int cnt = (int)min((long long)KNN_TILE, nr - tile);
+      __syncthreads();
+      for (int i = threadIdx.x; i < cnt * F; i += blockDim.x)
+        s_r[i] = R[tile * F + i];
+      if (ry)
+        for (int i = threadIdx.x; i < cnt; i += blockDim.x) s_y[i] = ry[tile + i];
+      __syncthreads();
+      if (q >= nq) continue;
+      for (int s = 0; s < cnt; ++s) {
+        float d = 0.f;
+#pragma unroll
+        for (int j = 0; j < F; ++j) {
+          float t = x.v[j] - s_r[s * F + j];
+          d = fmaf(t, t, d);
+        }
+        if (d < bd[K - 1]) {
+          // branchless sorted insert: slot k takes bd[k-1] when the new
+          // element lands above it, or the new element when it lands here;
+          // strict compares keep earlier (lower) reference indices on ties.
+          int ins = (int)(tile + s);
+          int lab = ry ? (int)s_y[s] : 0;
+#pragma unroll
+          for (int k = K - 1; k > 0; --k) {
+            bool above = bd[k - 1] > d;  // new element goes before slot k-1
+            if (bd[k] > d) {
+              bd[k] = above ? bd[k - 1] : d;
+              bi_[k] = above ? bi_[k - 1] : ins;
+              bl[k] = above ? bl[k - 1] : lab;
+            }
+          }
+          if (bd[0] > d) { bd[0] = d; bi_[0] = ins; bl[0] = lab; }
+        }
+      }
+    }
+    if (q >= nq) continue;
+#pragma unroll
+    for (int k = 0; k < K; ++k) {
+      out_d[q * K + k] = bd[k];
+      out_i[q * K + k] = bi_[k] >= 0 ? (int)(bi_[k] + idx_base) : -1;
+    }
+    if (out_lab) {
+      int best = 0, bc = 0;
+      for (int c = 0; c < C; ++c) {
+        int v = 0;
+#pragma unroll
+        for (int k = 0; k < K; ++k) v += (bl[k] == c && bi_[k] >= 0) ? 1 : 0;
+        if (v > best) { best = v; bc = c; }
+      }
+      out_lab[q] = bc;
+    }
+  }
+}
+
+extern "C" void launch_knn_topk(const float* Q, const float* R,
+                                const unsigned char* ry, float* out_d,
+                                int* out_i, int* out_lab, long long nq,
+                                long long nr, int k, int C, long long idx_base,
+                                hipStream_t stream) {
+  const int block = 256;
+  dim3 grid(ts_grid(nq, block));
+#define KNN_CASE(KV)                                                        \
+  case KV:                                                                  \
+    hipLaunchKernelGGL((knn_topk_kernel<KV>), grid, dim3(block), 0, stream, \
+                       Q, R, ry, out_d, out_i, out_lab, nq, nr, C, idx_base); \
+    return;
+  switch (k) {
+    KNN_CASE(1) KNN_CASE(2) KNN_CASE(3) KNN_CASE(4) KNN_CASE(5) KNN_CASE(6)
+    KNN_CASE(7) KNN_CASE(8) KNN_CASE(16) KNN_CASE(32)
+    default: break;
+  }
+#undef KNN_CASE
+}

@@ -120,8 +120,7 @@ class KMeans(Estimator):
 
     def predict_index(self, X: ArrayLike) -> torch.Tensor:
         Xt = as_tensor(X, self.device, torch.float32)
-        labels, _, _, _ = ops.kmeans_assign(Xt, self.cluster_centers_.to(Xt.dtype))
-        return labels
+        return ops.kmeans_labels(Xt, self.cluster_centers_.to(Xt.dtype))
 
     # -- checkpointing -------------------------------------------------
     def to_params(self) -> Dict[str, Any]:

@@ -133,6 +133,15 @@ def knn_topk(Q: torch.Tensor, R: torch.Tensor, k: int) -> Tuple[torch.Tensor, to
     return dist, idx
 
 
+def knn_classify(
+    Q: torch.Tensor, R: torch.Tensor, y: torch.Tensor, k: int, n_classes: int, idx_base: int = 0
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Fused top-k + uniform vote (CPU oracle for the fused GPU kernel)."""
+    dist, idx = knn_topk(Q, R, k)
+    lab = knn_vote(idx, y.long(), n_classes)
+    return dist, idx + idx_base, lab
+
+
 def knn_vote(idx: torch.Tensor, y: torch.Tensor, n_classes: int) -> torch.Tensor:
     """Uniform-weight majority vote; ties -> lowest class index (sklearn mode)."""
     labels = y[idx]  # (nq, k)
@@ -308,3 +317,51 @@ def rf_predict_proba(X: torch.Tensor, forest: Dict[str, torch.Tensor]) -> torch.
 
 def rf_argmax(X: torch.Tensor, forest: Dict[str, torch.Tensor]) -> torch.Tensor:
     return torch.argmax(rf_predict_proba(X, forest), dim=1).to(torch.int32)
+
+
+# ----------------------------------------------------------------------
+# Serve-path feature extraction (CPU oracle of the GPU kernel; mirrors the
+# reference Flow update math, traffic_classifier.py:63-96)
+# ----------------------------------------------------------------------
+
+
+def flow_features(cur: torch.Tensor, prev: torch.Tensor, times: torch.Tensor) -> torch.Tensor:
+    """counters -> 12-feature rows.
+
+    cur/prev: (n,4) cumulative [fwd_pkts, fwd_bytes, rev_pkts, rev_bytes];
+    times: (n,4) [t_now, t_prev_fwd, t_prev_rev, t_start].
+    Division guards: a zero time delta leaves the rate at 0 (the reference
+    skips the update; with prev==cur at creation the result matches).
+    """
+    d = cur - prev
+    t = times[:, 0]
+    life = t - times[:, 3]
+    df = t - times[:, 1]
+    dr = t - times[:, 2]
+
+    def safe(num, den):
+        out = torch.zeros_like(num)
+        nz = den != 0
+        out[nz] = num[nz] / den[nz]
+        return out
+
+    cols = [
+        d[:, 0],
+        d[:, 1],
+        safe(d[:, 0], df),
+        safe(cur[:, 0], life),
+        safe(d[:, 1], df),
+        safe(cur[:, 1], life),
+        d[:, 2],
+        d[:, 3],
+        safe(d[:, 2], dr),
+        safe(cur[:, 2], life),
+        safe(d[:, 3], dr),
+        safe(cur[:, 3], life),
+    ]
+    return torch.stack(cols, dim=1).to(torch.float32)
+
+
+def kmeans_labels(X: torch.Tensor, centers: torch.Tensor) -> torch.Tensor:
+    """Assignment only (predict path; no partial-sum update)."""
+    return kmeans_assign(X, centers)[0]

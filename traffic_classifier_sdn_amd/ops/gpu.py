@@ -1,0 +1,182 @@
+"""GPU op bindings: hand-written CDNA4 HIP kernels (csrc/*.hip).
+
+Every op the models call on CUDA tensors is bound here explicitly.  Hot ops
+run the HIP kernels; a handful of cold diagnostic ops (decision functions,
+seeding distances) deliberately alias the torch implementations — they are
+not part of any benched path.  Import fails loudly when the extension is
+missing (see ops.__init__)."""
+
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import torch
+
+from . import cpu as _cpu
+from . import _tcsdn_hip as _ext  # type: ignore  # built by setup.py
+
+# cold / diagnostic aliases (torch ops, not benched)
+linear_logits = _cpu.linear_logits
+gnb_joint_loglik = _cpu.gnb_joint_loglik
+pairwise_sqdist = _cpu.pairwise_sqdist
+rbf_kernel = _cpu.rbf_kernel
+svc_ovo_decision = _cpu.svc_ovo_decision
+svc_vote = _cpu.svc_vote
+knn_vote = _cpu.knn_vote
+rf_flatten = _cpu.rf_flatten
+
+
+def _f32(t: torch.Tensor) -> torch.Tensor:
+    return t.to(torch.float32).contiguous()
+
+
+def _f64(t: torch.Tensor) -> torch.Tensor:
+    return t.to(torch.float64).contiguous()
+
+
+# ----------------------------------------------------------------------
+# predict ops
+# ----------------------------------------------------------------------
+
+
+def linear_argmax(X: torch.Tensor, coef: torch.Tensor, intercept: torch.Tensor) -> torch.Tensor:
+    return _ext.linear_argmax(_f32(X), _f32(coef), _f32(intercept))
+
+
+def gnb_argmax(
+    X: torch.Tensor, theta: torch.Tensor, var: torch.Tensor, class_prior: torch.Tensor
+) -> torch.Tensor:
+    # precompute per-class constants in f64 for stability, then cast
+    var64 = var.double()
+    const = (torch.log(class_prior.double()) - 0.5 * torch.log(2.0 * torch.pi * var64).sum(dim=1))
+    inv_var = (1.0 / var64).float().contiguous()
+    return _ext.gnb_predict(_f32(X), _f32(theta), inv_var, const.float().contiguous())
+
+
+def kmeans_assign(
+    X: torch.Tensor, centers: torch.Tensor
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+    labels, counts, sums, inertia = _ext.kmeans_assign(_f32(X), _f32(centers), True)
+    dt = X.dtype if X.dtype in (torch.float32, torch.float64) else torch.float32
+    return labels, counts.to(dt), sums.to(dt), inertia[0].to(dt)
+
+
+def knn_topk(Q: torch.Tensor, R: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor]:
+    dist, idx = _ext.knn_topk(_f32(Q), _f32(R), None, k, 0, 0)
+    return dist, idx
+
+
+def knn_classify(
+    Q: torch.Tensor, R: torch.Tensor, y: torch.Tensor, k: int, n_classes: int, idx_base: int = 0
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Fused top-k + uniform vote (labels also returned for sharded merge)."""
+    y8 = y.to(torch.uint8).contiguous()
+    dist, idx, lab = _ext.knn_topk(_f32(Q), _f32(R), y8, k, n_classes, idx_base)
+    return dist, idx, lab
+
+
+def svc_predict(
+    X: torch.Tensor,
+    SV: torch.Tensor,
+    dual_coef: torch.Tensor,
+    intercept: torch.Tensor,
+    n_support: torch.Tensor,
+    gamma: float,
+) -> torch.Tensor:
+    svclass = torch.repeat_interleave(
+        torch.arange(n_support.numel(), device=X.device), n_support.to(X.device)
+    ).to(torch.uint8)
+    return _ext.svc_predict(
+        _f32(X), _f32(SV), _f32(dual_coef), svclass.contiguous(), _f32(intercept), float(gamma)
+    )
+
+
+def rf_pack(forest: Dict[str, torch.Tensor], device) -> Dict[str, torch.Tensor]:
+    """Pack the SoA forest (ops.cpu.rf_flatten layout) into the traversal
+    kernel's uint2 node format:
+      word0 = f32 threshold bits (inner) | leaf-probability row (leaf)
+      word1 = (right_child_GLOBAL << 8) | feature, feature 0xff marks a leaf
+    """
+    thr = forest["threshold"].cpu()
+    right = forest["right"].cpu().to(torch.int64)
+    feat = forest["feature"].cpu().to(torch.int64)
+    leaf_index = forest["leaf_index"].cpu().to(torch.int64)
+    offsets = forest["tree_offset"].cpu().to(torch.int64)
+    n_nodes = thr.numel()
+    n_trees = offsets.numel() - 1
+    base = torch.repeat_interleave(offsets[:-1], offsets[1:] - offsets[:-1])
+    is_leaf = feat < 0
+    w0 = torch.where(
+        is_leaf, leaf_index.to(torch.int32), thr.view(torch.int32)
+    ).to(torch.int32)
+    right_global = (right + base) << 8
+    w1 = torch.where(
+        is_leaf,
+        torch.tensor(0xFF, dtype=torch.int64),
+        right_global | feat,
+    ).to(torch.int32)
+    nodes = torch.stack([w0, w1], dim=1).contiguous()
+    return {
+        "nodes": nodes.to(device),
+        "roots": offsets[:-1].to(torch.int32).to(device),
+        "leaf_proba": forest["leaf_proba"].to(torch.float32).to(device).contiguous(),
+        "n_leaves": int(forest["leaf_proba"].shape[0]),
+        "n_classes": int(forest["n_classes"]),
+    }
+
+
+def rf_argmax(X: torch.Tensor, forest: Dict[str, torch.Tensor]) -> torch.Tensor:
+    packed = forest.get("_packed")
+    if packed is None or packed["nodes"].device != X.device:
+        packed = rf_pack(forest, X.device)
+        forest["_packed"] = packed
+    return _ext.rf_predict(
+        _f32(X),
+        packed["nodes"],
+        packed["roots"],
+        packed["leaf_proba"],
+        packed["n_leaves"],
+        packed["n_classes"],
+    )
+
+
+# ----------------------------------------------------------------------
+# fit ops
+# ----------------------------------------------------------------------
+
+
+def gnb_fit_stats(
+    X: torch.Tensor, y: torch.Tensor, n_classes: int
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    count, s, sq = _ext.gnb_fit_stats(_f64(X), y.to(torch.int64).contiguous(), n_classes)
+    return count.to(X.dtype), s.to(X.dtype), sq.to(X.dtype)
+
+
+def logistic_loss_grad(
+    X: torch.Tensor,
+    y: torch.Tensor,
+    coef: torch.Tensor,
+    intercept: torch.Tensor,
+    l2: float = 1.0,
+    sample_range=None,
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    grad, loss = _ext.logistic_grad(
+        _f64(X), y.to(torch.int64).contiguous(), _f64(coef), _f64(intercept)
+    )
+    loss = loss[0] + 0.5 * l2 * (coef.double() * coef.double()).sum()
+    g_coef = grad[:, :12] + l2 * coef.double()
+    g_b = grad[:, 12]
+    return loss.to(X.dtype), g_coef.to(X.dtype), g_b.to(X.dtype)
+
+
+def flow_features(cur: torch.Tensor, prev: torch.Tensor, times: torch.Tensor) -> torch.Tensor:
+    return _ext.flow_features(_f64(cur), _f64(prev), _f64(times))
+
+
+def kmeans_labels(X: torch.Tensor, centers: torch.Tensor) -> torch.Tensor:
+    labels, _, _, _ = _ext.kmeans_assign(_f32(X), _f32(centers), False)
+    return labels
+
+
+# diagnostic-only path (torch eager; not benched)
+rf_predict_proba = _cpu.rf_predict_proba

@@ -173,7 +173,7 @@ static std::vector<torch::Tensor> gnb_fit_stats(torch::Tensor X,
   auto count = torch::zeros({C}, X.options());
   auto sum = torch::zeros({C, 12}, X.options());
   auto sumsq = torch::zeros({C, 12}, X.options());
-  launch_gnb_fit_stats(X.data_ptr<double>(), y.data_ptr<long long>(),
+  launch_gnb_fit_stats(X.data_ptr<double>(), reinterpret_cast<const long long*>(y.data_ptr<int64_t>()),
                        count.data_ptr<double>(), sum.data_ptr<double>(),
                        sumsq.data_ptr<double>(), n, (int)C, cur_stream());
   return {count, sum, sumsq};
@@ -191,7 +191,7 @@ static std::vector<torch::Tensor> logistic_grad(torch::Tensor X,
   const int C = W.size(0);
   auto grad = torch::zeros({C, 13}, X.options());
   auto loss = torch::zeros({1}, X.options());
-  launch_logistic_grad(X.data_ptr<double>(), y.data_ptr<long long>(),
+  launch_logistic_grad(X.data_ptr<double>(), reinterpret_cast<const long long*>(y.data_ptr<int64_t>()),
                        W.data_ptr<double>(), b.data_ptr<double>(),
                        grad.data_ptr<double>(), loss.data_ptr<double>(), n, C,
                        cur_stream());

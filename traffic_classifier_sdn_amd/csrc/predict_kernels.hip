@@ -185,66 +185,99 @@ extern "C" void launch_kmeans_assign(const float* X, const float* centers,
 // Random forest traversal + vote (N4) — the flagship predict kernel.
 // Packed node = uint2 { x: f32 threshold bits | leaf-prob row index,
 //                       y: (right_child_global << 8) | feature (0xff = leaf) }
-// The whole packed forest (and optionally the leaf-probability table) is
-// staged in LDS; each lane walks all trees for its row, accumulating the
-// leaf class distributions, then writes argmax.  The per-lane feature row
-// lives in LDS (node feature ids are runtime indices).
+// Design (profiled on MI355X): the traversal is a dependent chain of node
+// fetches, so occupancy is the lever.  Only the packed forest lives in LDS
+// (shared by the whole block); the per-lane feature row stays in REGISTERS
+// and the runtime feature index is resolved with an 11-op cndmask select
+// tree (compile-time indices, ~8 cycles — far shorter than an LDS
+// round-trip and no scratch spill).  512-thread blocks with nodes-only LDS
+// give 3 blocks = 24 waves per CU.
 // ---------------------------------------------------------------------------
-template <int C>
+
+// select v[f] for runtime f in [0,12) with compile-time register indices:
+// two levels — pick r = f&3 within each quad, then the quad q = f>>2.
+DEV float sel12(const Row12& x, unsigned f) {
+  unsigned r = f & 3u, q = f >> 2;
+  float a = (r & 1u) ? x.v[1] : x.v[0];
+  float b = (r & 2u) ? ((r & 1u) ? x.v[3] : x.v[2]) : a;
+  float q0 = b;
+  float c = (r & 1u) ? x.v[5] : x.v[4];
+  float d = (r & 2u) ? ((r & 1u) ? x.v[7] : x.v[6]) : c;
+  float q1 = d;
+  float e = (r & 1u) ? x.v[9] : x.v[8];
+  float g = (r & 2u) ? ((r & 1u) ? x.v[11] : x.v[10]) : e;
+  float q2 = g;
+  float lo = (q & 1u) ? q1 : q0;
+  return (q & 2u) ? q2 : lo;
+}
+
+// Traversal body shared by the four table-placement variants.  LDS_NODES /
+// LDS_PROBS are compile-time so the node fetch lowers to ds_read_b64 (LDS)
+// or global_load_dwordx2 — a runtime ternary over the two pointers would
+// force generic/flat addressing (measured: LDS instr count collapses and
+// VALU doubles on 64-bit flat address math).
+//
+// Leaf encoding: most leaves of a fully-grown forest are PURE (one-hot
+// distribution).  A pure leaf carries its class in the feature byte
+// (0xf0|class) and is counted with one shift+add into a packed u64 vote
+// register (10 bits per class, forests up to 1023 trees) — no memory read
+// at all.  Only mixed leaves (feature byte 0xff) touch the probability
+// table.
+template <int C, bool LDS_NODES, bool LDS_PROBS>
+__launch_bounds__(512, 1)
 __global__ void rf_predict_kernel(const float* __restrict__ X,
                                   const uint2* __restrict__ nodes,
                                   const int* __restrict__ roots,  // [T]
                                   const float* __restrict__ leaf_proba,
                                   int* __restrict__ out,
                                   long long n, int n_nodes, int n_leaves,
-                                  int T, int lds_nodes, int lds_probs) {
-  constexpr int F = 12;
+                                  int T) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: [row features 256*F f32][nodes uint2][leaf probs f32]
-  float* s_x = reinterpret_cast<float*>(smem);  // [block][F]
-  uint2* s_nodes = reinterpret_cast<uint2*>(s_x + blockDim.x * F);
-  float* s_probs = reinterpret_cast<float*>(s_nodes + (lds_nodes ? n_nodes : 0));
+  uint2* s_nodes = reinterpret_cast<uint2*>(smem);
+  float* s_probs = reinterpret_cast<float*>(s_nodes + (LDS_NODES ? n_nodes : 0));
 
-  if (lds_nodes)
+  if (LDS_NODES)
     for (int i = threadIdx.x; i < n_nodes; i += blockDim.x) s_nodes[i] = nodes[i];
-  if (lds_probs)
+  if (LDS_PROBS)
     for (int i = threadIdx.x; i < n_leaves * C; i += blockDim.x)
       s_probs[i] = leaf_proba[i];
-  __syncthreads();
-
-  const uint2* nd = lds_nodes ? s_nodes : nodes;
-  const float* lp = lds_probs ? s_probs : leaf_proba;
-  float* my_x = s_x + threadIdx.x * F;
+  if (LDS_NODES || LDS_PROBS) __syncthreads();
 
   long long stride = (long long)gridDim.x * blockDim.x;
   for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x; row < n;
        row += stride) {
     Row12 x = load_row12(X, row);
-#pragma unroll
-    for (int j = 0; j < F; ++j) my_x[j] = x.v[j];
     float acc[C];
 #pragma unroll
     for (int c = 0; c < C; ++c) acc[c] = 0.f;
+    unsigned long long votes = 0ull;  // 10-bit packed per-class pure counts
     for (int t = 0; t < T; ++t) {
       int idx = roots[t];
       while (true) {
-        uint2 node = nd[idx];
+        uint2 node = LDS_NODES ? s_nodes[idx] : nodes[idx];
         unsigned feat = node.y & 0xffu;
-        if (feat == 0xffu) {
-          const float* p = lp + (long long)node.x * C;
+        if (feat >= 0xf0u) {
+          if (feat == 0xffu) {  // mixed leaf: read the distribution
+            int pr = (int)node.x * C;
 #pragma unroll
-          for (int c = 0; c < C; ++c) acc[c] += p[c];
+            for (int c = 0; c < C; ++c)
+              acc[c] += LDS_PROBS ? s_probs[pr + c] : leaf_proba[pr + c];
+          } else {  // pure leaf: class in the low nibble
+            votes += 1ull << ((feat & 0xfu) * 10);
+          }
           break;
         }
         float thr = __uint_as_float(node.x);
-        idx = (my_x[feat] <= thr) ? idx + 1 : (int)(node.y >> 8);
+        idx = (sel12(x, feat) <= thr) ? idx + 1 : (int)(node.y >> 8);
       }
     }
     float best = -INFINITY;
     int bi = 0;
 #pragma unroll
-    for (int c = 0; c < C; ++c)
-      if (acc[c] > best) { best = acc[c]; bi = c; }
+    for (int c = 0; c < C; ++c) {
+      float sc = acc[c] + (float)((votes >> (c * 10)) & 1023ull);
+      if (sc > best) { best = sc; bi = c; }
+    }
     out[row] = bi;
   }
 }
@@ -254,23 +287,27 @@ extern "C" void launch_rf_predict(const float* X, const unsigned* nodes,
                                   int* out, long long n, int n_nodes,
                                   int n_leaves, int T, int C,
                                   hipStream_t stream) {
-  const int block = 256;
-  // LDS budget: keep under 128 KiB so the row-feature slab plus forest fit;
-  // fall back to global (L2-resident) tables for big forests.
-  size_t row_bytes = (size_t)block * 12 * sizeof(float);
+  const int block = 512;
+  // Occupancy first: with nodes-only LDS a 43 KB forest admits 3 blocks
+  // (24 waves) per CU.  Leaf probabilities go to LDS only when everything
+  // still fits the 3-block budget; big forests fall back to the
+  // (L2-resident) global tables.
   size_t node_bytes = (size_t)n_nodes * sizeof(uint2);
   size_t prob_bytes = (size_t)n_leaves * C * sizeof(float);
-  size_t budget = 128 * 1024;
-  int lds_nodes = (row_bytes + node_bytes) <= budget;
-  int lds_probs = lds_nodes && (row_bytes + node_bytes + prob_bytes) <= budget;
-  size_t lds = row_bytes + (lds_nodes ? node_bytes : 0) + (lds_probs ? prob_bytes : 0);
-  dim3 grid(ts_grid(n, block));
+  size_t budget = 52 * 1024;  // 3 blocks/CU floor
+  bool lds_nodes = node_bytes <= budget;
+  bool lds_probs = lds_nodes && (node_bytes + prob_bytes) <= budget;
+  size_t lds = (lds_nodes ? node_bytes : 0) + (lds_probs ? prob_bytes : 0);
+  dim3 grid(ts_grid(n, block, 4096));
+#define RF_LAUNCH(CV, LN, LP)                                                \
+  hipLaunchKernelGGL((rf_predict_kernel<CV, LN, LP>), grid, dim3(block),     \
+                     lds, stream, X, reinterpret_cast<const uint2*>(nodes),  \
+                     roots, leaf_proba, out, n, n_nodes, n_leaves, T)
 #define RF_CASE(CV)                                                          \
   case CV:                                                                   \
-    hipLaunchKernelGGL((rf_predict_kernel<CV>), grid, dim3(block), lds,      \
-                       stream, X, reinterpret_cast<const uint2*>(nodes),     \
-                       roots, leaf_proba, out, n, n_nodes, n_leaves, T,      \
-                       lds_nodes, lds_probs);                                \
+    if (lds_nodes && lds_probs) RF_LAUNCH(CV, true, true);                   \
+    else if (lds_nodes) RF_LAUNCH(CV, true, false);                          \
+    else RF_LAUNCH(CV, false, false);                                        \
     return;
   switch (C) {
     RF_CASE(2) RF_CASE(3) RF_CASE(4) RF_CASE(5) RF_CASE(6) RF_CASE(7)
@@ -278,6 +315,7 @@ extern "C" void launch_rf_predict(const float* X, const unsigned* nodes,
     default: break;
   }
 #undef RF_CASE
+#undef RF_LAUNCH
 }
 
 // ---------------------------------------------------------------------------

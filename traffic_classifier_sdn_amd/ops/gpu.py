@@ -94,8 +94,12 @@ def svc_predict(
 def rf_pack(forest: Dict[str, torch.Tensor], device) -> Dict[str, torch.Tensor]:
     """Pack the SoA forest (ops.cpu.rf_flatten layout) into the traversal
     kernel's uint2 node format:
-      word0 = f32 threshold bits (inner) | leaf-probability row (leaf)
-      word1 = (right_child_GLOBAL << 8) | feature, feature 0xff marks a leaf
+      word0 = f32 threshold bits (inner) | leaf-probability row (mixed leaf)
+      word1 = (right_child_GLOBAL << 8) | feature byte
+    Feature byte: 0..0xef = inner-node feature id; 0xf0|class = PURE leaf
+    (one-hot distribution, counted in-register by the kernel); 0xff = mixed
+    leaf (distribution row in word0).  Forests up to 1023 trees and 16
+    classes; features up to 0xef.
     """
     thr = forest["threshold"].cpu()
     right = forest["right"].cpu().to(torch.int64)
@@ -106,15 +110,20 @@ def rf_pack(forest: Dict[str, torch.Tensor], device) -> Dict[str, torch.Tensor]:
     n_trees = offsets.numel() - 1
     base = torch.repeat_interleave(offsets[:-1], offsets[1:] - offsets[:-1])
     is_leaf = feat < 0
+    probs = forest["leaf_proba"]
+    # pure leaf: exactly one class carries the whole mass
+    pmax, pcls = probs.max(dim=1)
+    leaf_pure = torch.zeros_like(is_leaf)
+    leaf_cls = torch.zeros_like(feat)
+    valid = leaf_index >= 0
+    leaf_pure[valid] = pmax[leaf_index[valid]] >= 1.0
+    leaf_cls[valid] = pcls[leaf_index[valid]].to(torch.int64)
     w0 = torch.where(
         is_leaf, leaf_index.to(torch.int32), thr.view(torch.int32)
     ).to(torch.int32)
     right_global = (right + base) << 8
-    w1 = torch.where(
-        is_leaf,
-        torch.tensor(0xFF, dtype=torch.int64),
-        right_global | feat,
-    ).to(torch.int32)
+    leaf_byte = torch.where(leaf_pure, 0xF0 | leaf_cls, torch.tensor(0xFF, dtype=torch.int64))
+    w1 = torch.where(is_leaf, leaf_byte, right_global | feat).to(torch.int32)
     nodes = torch.stack([w0, w1], dim=1).contiguous()
     return {
         "nodes": nodes.to(device),

@@ -155,7 +155,7 @@ extern "C" void launch_logistic_grad(const double* X, const long long* y,
 // ---------------------------------------------------------------------------
 __global__ void flow_features_kernel(const double* __restrict__ cur,   // [n,4] fp,fb,rp,rb
                                      const double* __restrict__ prev,  // [n,4]
-                                     const double* __restrict__ times, // [n,4] t,tf,tr,t0
+                                     const double* __restrict__ times, // [n,6] tfc,tfp,trc,trp,t0,_
                                      float* __restrict__ out,          // [n,12]
                                      long long n) {
   long long stride = (long long)gridDim.x * blockDim.x;
@@ -163,26 +163,26 @@ __global__ void flow_features_kernel(const double* __restrict__ cur,   // [n,4] 
        i += stride) {
     double fp = cur[i * 4 + 0], fb = cur[i * 4 + 1];
     double rp = cur[i * 4 + 2], rb = cur[i * 4 + 3];
-    double pfp = prev[i * 4 + 0], pfb = prev[i * 4 + 1];
-    double prp = prev[i * 4 + 2], prb = prev[i * 4 + 3];
-    double t = times[i * 4 + 0], tf = times[i * 4 + 1];
-    double tr = times[i * 4 + 2], t0 = times[i * 4 + 3];
-    double dfp = fp - pfp, dfb = fb - pfb;
-    double drp = rp - prp, drb = rb - prb;
-    double life = t - t0, df = t - tf, dr = t - tr;
+    double dfp = fp - prev[i * 4 + 0], dfb = fb - prev[i * 4 + 1];
+    double drp = rp - prev[i * 4 + 2], drb = rb - prev[i * 4 + 3];
+    double tfc = times[i * 6 + 0], tfp = times[i * 6 + 1];
+    double trc = times[i * 6 + 2], trp = times[i * 6 + 3];
+    double t0 = times[i * 6 + 4];
+    double df = tfc - tfp, dr = trc - trp;
+    double lf = tfc - t0, lr = trc - t0;
     float* o = out + i * 12;
     o[0] = (float)dfp;
     o[1] = (float)dfb;
     o[2] = (float)(df != 0.0 ? dfp / df : 0.0);
-    o[3] = (float)(life != 0.0 ? fp / life : 0.0);
+    o[3] = (float)(lf != 0.0 ? fp / lf : 0.0);
     o[4] = (float)(df != 0.0 ? dfb / df : 0.0);
-    o[5] = (float)(life != 0.0 ? fb / life : 0.0);
+    o[5] = (float)(lf != 0.0 ? fb / lf : 0.0);
     o[6] = (float)drp;
     o[7] = (float)drb;
     o[8] = (float)(dr != 0.0 ? drp / dr : 0.0);
-    o[9] = (float)(life != 0.0 ? rp / life : 0.0);
+    o[9] = (float)(lr != 0.0 ? rp / lr : 0.0);
     o[10] = (float)(dr != 0.0 ? drb / dr : 0.0);
-    o[11] = (float)(life != 0.0 ? rb / life : 0.0);
+    o[11] = (float)(lr != 0.0 ? rb / lr : 0.0);
   }
 }
 

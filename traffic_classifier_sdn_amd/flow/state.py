@@ -56,7 +56,16 @@ _R_AVG_BPS = 17
 _R_LAST_TIME = 18
 _F_ACTIVE = 19
 _R_ACTIVE = 20
-_STATE_COLS = 21
+# raw counter snapshots for the GPU feature-extraction path: previous
+# cumulative counters and the update times bracketing the last change of
+# each direction (serve_gpu hands these to the flow_features kernel)
+_F_PREV_PKTS = 21
+_F_PREV_BYTES = 22
+_F_PREV_TIME = 23
+_R_PREV_PKTS = 24
+_R_PREV_BYTES = 25
+_R_PREV_TIME = 26
+_STATE_COLS = 27
 
 # Feature-matrix column order == utils.schema.FEATURE_NAMES order.
 _FEATURE_COLS = np.array(
@@ -194,11 +203,20 @@ class FlowTable:
         # forward starts ACTIVE, reverse INACTIVE (traffic_classifier.py:47,59)
         s[_F_ACTIVE] = 1.0
         s[_R_ACTIVE] = 0.0
+        s[_F_PREV_PKTS] = packets
+        s[_F_PREV_BYTES] = bytes_
+        s[_F_PREV_TIME] = time
+        s[_R_PREV_PKTS] = 0.0
+        s[_R_PREV_BYTES] = 0.0
+        s[_R_PREV_TIME] = time
         return slot
 
     def _update_forward(self, slot: int, packets: int, bytes_: int, time: int) -> None:
         # reference: traffic_classifier.py:63-78
         s = self._state[slot]
+        s[_F_PREV_PKTS] = s[_F_PKTS]
+        s[_F_PREV_BYTES] = s[_F_BYTES]
+        s[_F_PREV_TIME] = s[_F_LAST_TIME]
         s[_F_DELTA_PKTS] = packets - s[_F_PKTS]
         s[_F_PKTS] = packets
         if time != s[_TIME_START]:
@@ -217,6 +235,9 @@ class FlowTable:
     def _update_reverse(self, slot: int, packets: int, bytes_: int, time: int) -> None:
         # reference: traffic_classifier.py:81-96
         s = self._state[slot]
+        s[_R_PREV_PKTS] = s[_R_PKTS]
+        s[_R_PREV_BYTES] = s[_R_BYTES]
+        s[_R_PREV_TIME] = s[_R_LAST_TIME]
         s[_R_DELTA_PKTS] = packets - s[_R_PKTS]
         s[_R_PKTS] = packets
         if time != s[_TIME_START]:
@@ -255,6 +276,24 @@ class FlowTable:
                 )
             )
         return out
+
+    def counters_snapshot(self):
+        """Raw per-flow counter state for the GPU feature-extraction kernel:
+        (cur[n,4], prev[n,4], times[n,6]) float64 arrays with layouts
+        cur/prev = [fwd_pkts, fwd_bytes, rev_pkts, rev_bytes],
+        times = [tf_cur, tf_prev, tr_cur, tr_prev, t_start, 0].
+        ops.flow_features(cur, prev, times) reproduces feature_matrix()
+        exactly (same math, same division guards)."""
+        st = self._state[: self._n]
+        cur = st[:, [_F_PKTS, _F_BYTES, _R_PKTS, _R_BYTES]].copy()
+        prev = st[:, [_F_PREV_PKTS, _F_PREV_BYTES, _R_PREV_PKTS, _R_PREV_BYTES]].copy()
+        times = np.zeros((self._n, 6), dtype=np.float64)
+        times[:, 0] = st[:, _F_LAST_TIME]
+        times[:, 1] = st[:, _F_PREV_TIME]
+        times[:, 2] = st[:, _R_LAST_TIME]
+        times[:, 3] = st[:, _R_PREV_TIME]
+        times[:, 4] = st[:, _TIME_START]
+        return cur, prev, times
 
     def metas(self) -> List[FlowMeta]:
         return list(self._meta)

@@ -326,18 +326,20 @@ def rf_argmax(X: torch.Tensor, forest: Dict[str, torch.Tensor]) -> torch.Tensor:
 
 
 def flow_features(cur: torch.Tensor, prev: torch.Tensor, times: torch.Tensor) -> torch.Tensor:
-    """counters -> 12-feature rows.
+    """counters -> 12-feature rows (mirrors Flow update math,
+    traffic_classifier.py:63-96).
 
-    cur/prev: (n,4) cumulative [fwd_pkts, fwd_bytes, rev_pkts, rev_bytes];
-    times: (n,4) [t_now, t_prev_fwd, t_prev_rev, t_start].
+    cur/prev: (n,4) cumulative [fwd_pkts, fwd_bytes, rev_pkts, rev_bytes] at
+    the latest / previous update of each direction;
+    times: (n,6) [tf_cur, tf_prev, tr_cur, tr_prev, t_start, pad].
     Division guards: a zero time delta leaves the rate at 0 (the reference
     skips the update; with prev==cur at creation the result matches).
     """
     d = cur - prev
-    t = times[:, 0]
-    life = t - times[:, 3]
-    df = t - times[:, 1]
-    dr = t - times[:, 2]
+    df = times[:, 0] - times[:, 1]
+    dr = times[:, 2] - times[:, 3]
+    lf = times[:, 0] - times[:, 4]
+    lr = times[:, 2] - times[:, 4]
 
     def safe(num, den):
         out = torch.zeros_like(num)
@@ -349,15 +351,15 @@ def flow_features(cur: torch.Tensor, prev: torch.Tensor, times: torch.Tensor) ->
         d[:, 0],
         d[:, 1],
         safe(d[:, 0], df),
-        safe(cur[:, 0], life),
+        safe(cur[:, 0], lf),
         safe(d[:, 1], df),
-        safe(cur[:, 1], life),
+        safe(cur[:, 1], lf),
         d[:, 2],
         d[:, 3],
         safe(d[:, 2], dr),
-        safe(cur[:, 2], life),
+        safe(cur[:, 2], lr),
         safe(d[:, 3], dr),
-        safe(cur[:, 3], life),
+        safe(cur[:, 3], lr),
     ]
     return torch.stack(cols, dim=1).to(torch.float32)
 

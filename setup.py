@@ -12,17 +12,12 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
+import pybind11
+from setuptools import Extension
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "traffic_classifier_sdn_amd", "csrc")
-
-
-class BuildInOps(BuildExtension):
-    def get_ext_filename(self, name):
-        # place the .so at traffic_classifier_sdn_amd/ops/_tcsdn_hip.so
-        return super().get_ext_filename(name)
-
 
 setup(
     name="traffic_classifier_sdn_amd",
@@ -40,7 +35,15 @@ setup(
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
             },
-        )
+        ),
+        # plain C++ extension: no torch/HIP dependency, loads standalone
+        Extension(
+            name="traffic_classifier_sdn_amd.flow._tcsdn_native",
+            sources=[os.path.join(CSRC, "flowtable.cpp")],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17"],
+            language="c++",
+        ),
     ],
     cmdclass={"build_ext": BuildExtension},
 )

@@ -1,0 +1,60 @@
+"""C++ flow table parity with the Python FlowTable on identical streams."""
+
+import numpy as np
+import pytest
+
+from traffic_classifier_sdn_amd.flow import native
+from traffic_classifier_sdn_amd.flow.parser import PollStreamParser
+from traffic_classifier_sdn_amd.flow.replay import TelemetryReplaySource
+
+pytestmark = pytest.mark.skipif(not native.HAVE_NATIVE, reason="native ext not built")
+
+
+def _both(polls=10, seed=5):
+    lines = list(TelemetryReplaySource(seed=seed).stream(polls))
+    p_py = PollStreamParser()
+    p_py.feed_many(lines)
+    p_nat = native.NativePollParser()
+    p_nat.feed_buffer("\n".join(lines) + "\n")
+    return p_py, p_nat
+
+
+def test_feature_matrix_parity():
+    p_py, p_nat = _both()
+    a = p_py.table.feature_matrix(dtype=np.float32)
+    b = p_nat.table.feature_matrix()
+    assert a.shape == b.shape
+    np.testing.assert_allclose(b, a, rtol=1e-7, atol=1e-7)
+    assert p_nat.records == p_py.records
+    assert len(p_nat.table) == len(p_py.table)
+
+
+def test_counters_snapshot_parity():
+    p_py, p_nat = _both(seed=6)
+    for x, y in zip(p_py.table.counters_snapshot(), p_nat.table.counters_snapshot()):
+        np.testing.assert_allclose(y, x)
+
+
+def test_statuses_and_metas():
+    p_py, p_nat = _both(seed=7)
+    assert p_nat.table.statuses() == p_py.table.statuses()
+    mp = [(m.datapath, m.inport, m.ethsrc, m.ethdst, m.outport) for m in p_py.table.metas()]
+    assert p_nat.table.metas() == mp
+
+
+def test_malformed_lines_counted():
+    p = native.NativePollParser()
+    assert p.feed("data\t100\t1") is None
+    assert p.feed("garbage") is None
+    assert p.feed("data\tx\t1\t1\ta\tb\t2\t1\t2") is None
+    assert p.bad_lines == 2
+    assert p.records == 0
+
+
+def test_bulk_vs_per_line():
+    lines = list(TelemetryReplaySource(seed=8).stream(5))
+    a = native.NativePollParser()
+    a.feed_many(lines)
+    b = native.NativePollParser()
+    b.feed_buffer("\n".join(lines))
+    np.testing.assert_allclose(b.table.feature_matrix(), a.table.feature_matrix())

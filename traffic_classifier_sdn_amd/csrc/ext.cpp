@@ -42,6 +42,10 @@ void launch_logistic_grad(const double*, const long long*, const double*,
                           hipStream_t);
 void launch_flow_features(const double*, const double*, const double*, float*,
                           long long, hipStream_t);
+void launch_smo_select(const float*, const double*, const double*, double,
+                       long long, unsigned long long*, hipStream_t);
+void launch_smo_update(const float*, const float*, double*, const float*,
+                       double, double, float, long long, hipStream_t);
 }
 
 static torch::Tensor gnb_predict(torch::Tensor X, torch::Tensor theta,
@@ -211,7 +215,33 @@ static torch::Tensor flow_features(torch::Tensor cur, torch::Tensor prev,
   return out;
 }
 
+static void smo_select(torch::Tensor y, torch::Tensor alpha,
+                       torch::Tensor grad, double C, torch::Tensor out) {
+  CHECK_IN(y, torch::kFloat32);
+  CHECK_IN(alpha, torch::kFloat64);
+  CHECK_IN(grad, torch::kFloat64);
+  CHECK_IN(out, torch::kInt64);  // reinterpreted as u64 packed keys, len 2
+  launch_smo_select(y.data_ptr<float>(), alpha.data_ptr<double>(),
+                    grad.data_ptr<double>(), C, y.size(0),
+                    reinterpret_cast<unsigned long long*>(out.data_ptr<int64_t>()),
+                    cur_stream());
+}
+
+static void smo_update(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
+                       torch::Tensor rows, double yidai, double yjdaj,
+                       double gamma) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(y, torch::kFloat32);
+  CHECK_IN(grad, torch::kFloat64);
+  CHECK_IN(rows, torch::kFloat32);
+  launch_smo_update(X.data_ptr<float>(), y.data_ptr<float>(),
+                    grad.data_ptr<double>(), rows.data_ptr<float>(), yidai,
+                    yjdaj, (float)gamma, X.size(0), cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("smo_select", &smo_select, "WSS-1 pair candidate selection");
+  m.def("smo_update", &smo_update, "fused RBF-row gradient update");
   m.def("gnb_predict", &gnb_predict, "fused GaussianNB loglik+argmax");
   m.def("linear_argmax", &linear_argmax, "logits+argmax");
   m.def("kmeans_assign", &kmeans_assign, "Lloyd assign + partial update");

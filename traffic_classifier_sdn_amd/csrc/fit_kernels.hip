@@ -193,3 +193,100 @@ extern "C" void launch_flow_features(const double* cur, const double* prev,
   hipLaunchKernelGGL(flow_features_kernel, dim3(ts_grid(n, block)), dim3(block),
                      0, stream, cur, prev, times, out, n);
 }
+
+// ---------------------------------------------------------------------------
+// SMO working-pair kernels for the scalable RBF-SVC dual fit (N2 fit path,
+// BASELINE config #3).  The host (models/svc_fit.py) drives libsvm WSS-1
+// iterations; rows are sharded across ranks and the pair candidates are
+// merged with one all-gather per iteration.
+// ---------------------------------------------------------------------------
+
+// order-preserving f64->u32 key (monotone: a<b  =>  enc(a)<enc(b))
+DEV unsigned enc_f32(float f) {
+  unsigned u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+
+// out[0] = packed argmax over I_up of (-y*grad); out[1] = packed argmax
+// over I_low of (+y*grad)  (i.e. argmin of -y*grad).  pack = key<<32 | idx.
+__global__ void smo_select_kernel(const float* __restrict__ y,
+                                  const double* __restrict__ alpha,
+                                  const double* __restrict__ grad,
+                                  double C, long long n,
+                                  unsigned long long* __restrict__ out) {
+  __shared__ unsigned long long s_up, s_low;
+  if (threadIdx.x == 0) { s_up = 0; s_low = 0; }
+  __syncthreads();
+  unsigned long long up = 0, low = 0;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < n;
+       t += stride) {
+    float yt = y[t];
+    double a = alpha[t];
+    float myg = (float)(-(double)yt * grad[t]);
+    bool in_up = (yt > 0.f && a < C) || (yt < 0.f && a > 0.0);
+    bool in_low = (yt > 0.f && a > 0.0) || (yt < 0.f && a < C);
+    if (in_up) {
+      unsigned long long p = ((unsigned long long)enc_f32(myg) << 32) | (unsigned)t;
+      up = p > up ? p : up;
+    }
+    if (in_low) {
+      unsigned long long p = ((unsigned long long)enc_f32(-myg) << 32) | (unsigned)t;
+      low = p > low ? p : low;
+    }
+  }
+  atomicMax(&s_up, up);
+  atomicMax(&s_low, low);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    atomicMax(&out[0], s_up);
+    atomicMax(&out[1], s_low);
+  }
+}
+
+extern "C" void launch_smo_select(const float* y, const double* alpha,
+                                  const double* grad, double C, long long n,
+                                  unsigned long long* out, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(smo_select_kernel, dim3(ts_grid(n, block)), dim3(block), 0,
+                     stream, y, alpha, grad, C, n, out);
+}
+
+// grad[t] += y_t * (yi*dai*K(xi,x_t) + yj*daj*K(xj,x_t)); the two RBF rows
+// are computed on the fly (fused — no kernel matrix is ever materialised).
+// xi/xj come from the 24-float staging buffer `rows` (filled by the host
+// each iteration; works for remote rows in the sharded fit).
+__global__ void smo_update_kernel(const float* __restrict__ X,
+                                  const float* __restrict__ y,
+                                  double* __restrict__ grad,
+                                  const float* __restrict__ rows,  // [24]
+                                  double yidai, double yjdaj, float gamma,
+                                  long long n) {
+  __shared__ float s_rows[24];
+  if (threadIdx.x < 24) s_rows[threadIdx.x] = rows[threadIdx.x];
+  __syncthreads();
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < n;
+       t += stride) {
+    Row12 x = load_row12(X, t);
+    float di = 0.f, dj = 0.f;
+#pragma unroll
+    for (int k = 0; k < 12; ++k) {
+      float a = x.v[k] - s_rows[k];
+      float b = x.v[k] - s_rows[12 + k];
+      di = fmaf(a, a, di);
+      dj = fmaf(b, b, dj);
+    }
+    double ki = (double)__expf(-gamma * di);
+    double kj = (double)__expf(-gamma * dj);
+    grad[t] += (double)y[t] * (yidai * ki + yjdaj * kj);
+  }
+}
+
+extern "C" void launch_smo_update(const float* X, const float* y, double* grad,
+                                  const float* rows, double yidai, double yjdaj,
+                                  float gamma, long long n, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(smo_update_kernel, dim3(ts_grid(n, block)), dim3(block), 0,
+                     stream, X, y, grad, rows, yidai, yjdaj, gamma, n);
+}

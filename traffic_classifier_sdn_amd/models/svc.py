@@ -5,10 +5,9 @@ predict: fused RBF-Gram + one-vs-one vote op on the libsvm checkpoint layout
 intercepts — SURVEY.md §2.3).
 
 fit: one-vs-one SMO (libsvm WSS-1 working-set selection, analytic pair
-update) over precomputed kernel blocks — the per-pair subproblems are a few
-thousand rows for the reference dataset.  Kernel-matrix tiles are computed by
-the distance-GEMM op (MFMA on GPU).  Distributed fit shards rows and
-all-reduces kernel-row partial sums (planned; single-process fit is complete).
+update) with on-the-fly fused kernel rows — no Gram matrix is materialised,
+so the fit scales to the BASELINE 1M-row config; rows shard across ranks
+with one candidate all-gather per iteration (models/svc_fit.py).
 """
 
 from __future__ import annotations
@@ -19,68 +18,9 @@ import numpy as np
 import torch
 
 from .. import ops
+from ..parallel import dist
 from .base import ArrayLike, Estimator, as_tensor, encode_labels
-
-
-def _smo_binary(K: torch.Tensor, y: torch.Tensor, C: float, tol: float, max_iter: int):
-    """libsvm-style SMO for one binary subproblem.
-
-    K: (n, n) kernel matrix (f64), y: (n,) in {-1, +1} (f64).
-    Returns (alpha, b).  Working-set selection is the maximal-violating-pair
-    rule; the pair update is the analytic two-variable solution.
-    """
-    n = K.shape[0]
-    alpha = torch.zeros(n, dtype=K.dtype, device=K.device)
-    grad = -torch.ones(n, dtype=K.dtype, device=K.device)  # G = Q@alpha - 1
-    Qdiag = torch.diagonal(K).clone()  # y_i^2 * K_ii
-    minus_yG = None
-    for _ in range(max_iter):
-        minus_yG = -y * grad
-        up = ((y > 0) & (alpha < C)) | ((y < 0) & (alpha > 0))
-        low = ((y > 0) & (alpha > 0)) | ((y < 0) & (alpha < C))
-        m_up = torch.where(up, minus_yG, torch.tensor(-np.inf, dtype=K.dtype, device=K.device))
-        m_low = torch.where(low, minus_yG, torch.tensor(np.inf, dtype=K.dtype, device=K.device))
-        i = int(torch.argmax(m_up))
-        j = int(torch.argmin(m_low))
-        if float(m_up[i]) - float(m_low[j]) < tol:
-            break
-        yi, yj = float(y[i]), float(y[j])
-        Qi = y[i] * y * K[i]  # row i of Q
-        Qj = y[j] * y * K[j]
-        a = float(Qdiag[i] + Qdiag[j] - 2.0 * yi * yj * K[i, j])
-        if a <= 0:
-            a = 1e-12
-        b_ = float(m_up[i] - m_low[j])
-        # step along the pair direction, then clip to the box
-        d = b_ / a
-        ai_old, aj_old = float(alpha[i]), float(alpha[j])
-        ai = ai_old + yi * d
-        aj = aj_old - yj * d
-        # clip: keep yi*ai + yj*aj constant
-        s = yi * ai_old + yj * aj_old
-        ai = min(max(ai, 0.0), C)
-        aj = yj * (s - yi * ai)
-        aj = min(max(aj, 0.0), C)
-        ai = yi * (s - yj * aj)
-        ai = min(max(ai, 0.0), C)
-        dai, daj = ai - ai_old, aj - aj_old
-        if abs(dai) < 1e-16 and abs(daj) < 1e-16:
-            break
-        alpha[i] = ai
-        alpha[j] = aj
-        grad += Qi * dai + Qj * daj
-    # intercept: average -y*G over free vectors, else midpoint of bounds
-    minus_yG = -y * grad
-    free = (alpha > 1e-12) & (alpha < C - 1e-12)
-    if bool(free.any()):
-        rho = -float(minus_yG[free].mean())
-    else:
-        up = ((y > 0) & (alpha < C)) | ((y < 0) & (alpha > 0))
-        low = ((y > 0) & (alpha > 0)) | ((y < 0) & (alpha < C))
-        hi = float(torch.where(up, minus_yG, torch.tensor(-np.inf, dtype=K.dtype)).max())
-        lo = float(torch.where(low, minus_yG, torch.tensor(np.inf, dtype=K.dtype)).min())
-        rho = -(hi + lo) / 2.0
-    return alpha, -rho  # decision uses +b with b = -rho
+from .svc_fit import smo_fit_pair
 
 
 class SVC(Estimator):
@@ -105,64 +45,91 @@ class SVC(Estimator):
         self.n_support_: Optional[torch.Tensor] = None
         self.gamma_: float = 0.0
 
-    def fit(self, X: ArrayLike, y: ArrayLike):
-        Xt = as_tensor(X, self.device, torch.float64)
-        self.classes_, y_idx = encode_labels(y)
+    def fit(self, X: ArrayLike, y: ArrayLike, sharded: bool = False):
+        """One-vs-one SMO fit (models/svc_fit.py: fused on-the-fly kernel
+        rows, HIP kernels on GPU, row-sharded across ranks when ``sharded``
+        with one candidate all-gather per iteration)."""
+        Xt = as_tensor(X, self.device, torch.float32)
+        n, F = Xt.shape
+        if sharded and dist.is_initialized():
+            classes_local = np.unique(np.asarray(y).ravel())
+            all_classes = [None] * dist.world_size()
+            torch.distributed.all_gather_object(all_classes, list(classes_local))
+            self.classes_ = np.unique(
+                np.concatenate([np.asarray(c) for c in all_classes])
+            ).astype(object)
+            lut = {c: i for i, c in enumerate(self.classes_)}
+            y_idx = torch.tensor([lut[v] for v in np.asarray(y).ravel()], dtype=torch.int64)
+        else:
+            self.classes_, y_idx = encode_labels(y)
         y_t = y_idx.to(self.device)
         C_cls = len(self.classes_)
-        n, F = Xt.shape
+
         if self.gamma == "scale":
-            xv = float(Xt.var(unbiased=False))
+            # sklearn: 1 / (F * Var(all elements of X)); global when sharded
+            st = torch.tensor(
+                [float(Xt.numel()), float(Xt.double().sum()), float((Xt.double() ** 2).sum())],
+                dtype=torch.float64,
+            )
+            dist.allreduce_(st)
+            xv = float(st[2] / st[0] - (st[1] / st[0]) ** 2)
             self.gamma_ = 1.0 / (F * xv) if xv > 0 else 1.0
         elif self.gamma == "auto":
             self.gamma_ = 1.0 / F
         else:
             self.gamma_ = float(self.gamma)
 
-        # per-pair SMO; remember alphas per (pair, global row)
-        alphas = {}
+        n_pairs = C_cls * (C_cls - 1) // 2
+        pair_alpha = torch.zeros(n, n_pairs, dtype=torch.float64)  # signed, local rows
         intercepts = []
+        p = 0
+        self.n_iter_ = []
         for i in range(C_cls):
             for j in range(i + 1, C_cls):
                 sel = (y_t == i) | (y_t == j)
                 idx = torch.nonzero(sel, as_tuple=False).squeeze(1)
-                Xp = Xt[idx]
-                yp = torch.where(y_t[idx] == i, 1.0, -1.0).to(Xt.dtype)
-                Kp = ops.rbf_kernel(Xp, Xp, self.gamma_)
-                a, b = _smo_binary(Kp, yp, self.C, self.tol, self.max_iter)
-                alphas[(i, j)] = (idx, a * yp)  # signed coefficients
+                Xp = Xt[idx].contiguous()
+                yp = torch.where(y_t[idx] == i, 1.0, -1.0).to(torch.float32)
+                alpha, b, iters = smo_fit_pair(
+                    Xp, yp, C=self.C, gamma=self.gamma_, tol=self.tol, max_iter=self.max_iter
+                )
+                pair_alpha[idx.cpu(), p] = (alpha.cpu() * yp.double().cpu())
                 intercepts.append(b)
+                self.n_iter_.append(iters)
+                p += 1
 
-        # assemble libsvm layout: SVs = rows with any nonzero coef, grouped
-        # by class in class order; dual_coef (C-1, nSV)
-        is_sv = torch.zeros(n, dtype=torch.bool)
-        for (i, j), (idx, sc) in alphas.items():
-            nz = sc.abs() > 1e-12
-            is_sv[idx[nz].cpu()] = True
-        order = []
-        n_support = []
-        for c in range(C_cls):
-            rows = torch.nonzero((y_t.cpu() == c) & is_sv, as_tuple=False).squeeze(1)
-            order.append(rows)
-            n_support.append(int(rows.numel()))
-        order_t = torch.cat(order)
-        nSV = int(order_t.numel())
-        pos_of = {int(r): p for p, r in enumerate(order_t.tolist())}
-        dual = torch.zeros(C_cls - 1, nSV, dtype=torch.float64)
-        for (i, j), (idx, sc) in alphas.items():
-            for local, g in enumerate(idx.cpu().tolist()):
-                v = float(sc[local])
-                if abs(v) <= 1e-12 or g not in pos_of:
-                    continue
-                c = int(y_t[g])  # class of this SV: i or j
-                other = j if c == i else i
-                row = other if other < c else other - 1
-                dual[row, pos_of[g]] = v
-        self.support_ = order_t.to(torch.int64)
-        self.support_vectors_ = Xt[order_t.to(Xt.device)].contiguous()
-        self.dual_coef_ = dual.to(self.device)
+        # assemble the libsvm layout (SVs grouped by class); gather shards
+        sv_mask = pair_alpha.abs().sum(dim=1) > 1e-12
+        Xsv_l = Xt[sv_mask.to(self.device)].double().cpu().numpy()
+        ysv_l = y_t[sv_mask.to(self.device)].cpu().numpy()
+        asv_l = pair_alpha[sv_mask].numpy()
+        if sharded and dist.is_initialized():
+            parts = [None] * dist.world_size()
+            torch.distributed.all_gather_object(parts, (Xsv_l, ysv_l, asv_l))
+            Xsv = np.concatenate([q[0] for q in parts])
+            ysv = np.concatenate([q[1] for q in parts])
+            asv = np.concatenate([q[2] for q in parts])
+        else:
+            Xsv, ysv, asv = Xsv_l, ysv_l, asv_l
+        order = np.argsort(ysv, kind="stable")
+        Xsv, ysv, asv = Xsv[order], ysv[order], asv[order]
+        nSV = Xsv.shape[0]
+        n_support = np.bincount(ysv, minlength=C_cls)
+        dual = np.zeros((C_cls - 1, nSV), dtype=np.float64)
+        p = 0
+        for i in range(C_cls):
+            for j in range(i + 1, C_cls):
+                col = asv[:, p]
+                for c, other in ((i, j), (j, i)):
+                    row = other if other < c else other - 1
+                    mask = ysv == c
+                    dual[row, mask] = col[mask]
+                p += 1
+        self.support_vectors_ = torch.as_tensor(Xsv).to(self.device)
+        self.dual_coef_ = torch.as_tensor(dual).to(self.device)
         self.intercept_ = torch.tensor(intercepts, dtype=torch.float64, device=self.device)
-        self.n_support_ = torch.tensor(n_support, dtype=torch.int64, device=self.device)
+        self.n_support_ = torch.as_tensor(n_support, dtype=torch.int64).to(self.device)
+        self.support_ = torch.nonzero(sv_mask, as_tuple=False).squeeze(1)
         return self
 
     def predict_index(self, X: ArrayLike) -> torch.Tensor:

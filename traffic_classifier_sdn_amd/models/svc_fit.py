@@ -1,0 +1,187 @@
+"""Scalable RBF-SVC dual fit: libsvm WSS-1 SMO with on-the-fly kernel rows.
+
+No kernel matrix is ever materialised — each iteration selects the maximal
+violating pair, solves the 2-variable subproblem analytically, and updates
+the dual gradient with ONE fused kernel pass over the local row shard
+(csrc smo_update: two RBF rows + gradient update in one sweep).  Rows are
+sharded across ranks (BASELINE config #3: 1M rows, 1/2/4/8 MI355X); per
+iteration the ranks exchange one small all-gather of pair candidates
+(value, row payload) — latency-bound, so everything is packed into a
+single collective.
+
+Device-generic: CUDA path uses the HIP kernels, CPU path the torch ops —
+identical control flow, so the multi-process logic is testable on gloo.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..parallel import dist
+
+
+def _local_select_cpu(y, alpha, grad, C):
+    myg = -(y.double() * grad)
+    up = ((y > 0) & (alpha < C)) | ((y < 0) & (alpha > 0))
+    low = ((y > 0) & (alpha > 0)) | ((y < 0) & (alpha < C))
+    ninf = torch.tensor(-math.inf, dtype=torch.float64)
+    pinf = torch.tensor(math.inf, dtype=torch.float64)
+    up_vals = torch.where(up, myg, ninf)
+    low_vals = torch.where(low, myg, pinf)
+    i = int(torch.argmax(up_vals))
+    j = int(torch.argmin(low_vals))
+    return i, float(up_vals[i]), j, float(low_vals[j])
+
+
+def _local_select_gpu(y, alpha, grad, C, _buf):
+    from ..ops import gpu as og
+
+    _buf.zero_()
+    og._ext.smo_select(y, alpha, grad, float(C), _buf)
+    packed = _buf.cpu().numpy().view(np.uint64)
+    i = int(packed[0] & 0xFFFFFFFF)
+    j = int(packed[1] & 0xFFFFFFFF)
+    if packed[0] == 0:
+        i = -1
+    if packed[1] == 0:
+        j = -1
+    up_val = float(-(float(y[i]) * float(grad[i]))) if i >= 0 else -math.inf
+    low_val = float(-(float(y[j]) * float(grad[j]))) if j >= 0 else math.inf
+    return i, up_val, j, low_val
+
+
+def _grad_update_cpu(X, y, grad, xi, xj, yidai, yjdaj, gamma):
+    di = ((X - xi) ** 2).sum(dim=1)
+    dj = ((X - xj) ** 2).sum(dim=1)
+    ki = torch.exp(-gamma * di).double()
+    kj = torch.exp(-gamma * dj).double()
+    grad += y.double() * (yidai * ki + yjdaj * kj)
+
+
+def smo_fit_pair(
+    X: torch.Tensor,
+    y_pm: torch.Tensor,
+    C: float = 1.0,
+    gamma: float = 1.0,
+    tol: float = 1e-3,
+    max_iter: int = 200_000,
+    check_every: int = 1,
+) -> Tuple[torch.Tensor, float, int]:
+    """SMO on the local shard (X[n_local,12] f32, y_pm in {-1,+1} f32).
+
+    Returns (alpha_local f64, intercept b, iterations).  With an initialised
+    process group the pair search is global (one fused all-gather/iter).
+    """
+    device = X.device
+    n = X.shape[0]
+    y = y_pm.to(torch.float32)
+    alpha = torch.zeros(n, dtype=torch.float64, device=device)
+    grad = -torch.ones(n, dtype=torch.float64, device=device)
+    is_gpu = device.type == "cuda"
+    sel_buf = torch.zeros(2, dtype=torch.int64, device=device) if is_gpu else None
+    rows_buf = torch.zeros(24, dtype=torch.float32, device=device) if is_gpu else None
+    world = dist.world_size()
+    it = 0
+    for it in range(1, max_iter + 1):
+        if is_gpu:
+            i, up_val, j, low_val = _local_select_gpu(y, alpha, grad, C, sel_buf)
+        else:
+            i, up_val, j, low_val = _local_select_cpu(y, alpha, grad, C)
+
+        # candidate payload: [up_val, low_val, xi(12), yi, ai, gi, xj(12), yj, aj, gj]
+        def payload(idx, val):
+            if idx < 0:
+                return [val] + [0.0] * 15
+            return (
+                [val]
+                + X[idx].double().tolist()
+                + [float(y[idx]), float(alpha[idx]), float(grad[idx])]
+            )
+
+        if world > 1:
+            local = torch.tensor(
+                payload(i, up_val) + payload(j, low_val), dtype=torch.float64
+            )
+            gathered = dist.allgather(local)
+            up_rank = int(np.argmax([float(g[0]) for g in gathered]))
+            low_rank = int(np.argmin([float(g[16]) for g in gathered]))
+            gu = gathered[up_rank]
+            gl = gathered[low_rank]
+            up_val = float(gu[0])
+            low_val = float(gl[16])
+            xi = gu[1:13]
+            yi, ai, gi = float(gu[13]), float(gu[14]), float(gu[15])
+            xj = gl[17:29]
+            yj, aj, gj = float(gl[29]), float(gl[30]), float(gl[31])
+            i_here = i if up_rank == dist.rank() else -1
+            j_here = j if low_rank == dist.rank() else -1
+        else:
+            if i < 0 or j < 0:
+                break
+            xi = X[i].double().cpu()
+            xj = X[j].double().cpu()
+            yi, ai, gi = float(y[i]), float(alpha[i]), float(grad[i])
+            yj, aj, gj = float(y[j]), float(alpha[j]), float(grad[j])
+            i_here, j_here = i, j
+
+        if up_val - low_val < tol:
+            break
+
+        # analytic 2-variable solve (K_ii = K_jj = 1 for RBF)
+        d2 = float(((xi - xj) ** 2).sum())
+        kij = math.exp(-gamma * d2)
+        a = 2.0 - 2.0 * yi * yj * kij
+        if a <= 0:
+            a = 1e-12
+        d = (up_val - low_val) / a
+        ai_new = ai + yi * d
+        aj_new = aj - yj * d
+        s = yi * ai + yj * aj
+        ai_new = min(max(ai_new, 0.0), C)
+        aj_new = yj * (s - yi * ai_new)
+        aj_new = min(max(aj_new, 0.0), C)
+        ai_new = yi * (s - yj * aj_new)
+        ai_new = min(max(ai_new, 0.0), C)
+        dai, daj = ai_new - ai, aj_new - aj
+        if abs(dai) < 1e-16 and abs(daj) < 1e-16:
+            break
+        if i_here >= 0:
+            alpha[i_here] = ai_new
+        if j_here >= 0:
+            alpha[j_here] = aj_new
+
+        xif = torch.as_tensor(xi, dtype=torch.float32)
+        xjf = torch.as_tensor(xj, dtype=torch.float32)
+        if is_gpu:
+            from ..ops import gpu as og
+
+            rows_buf[:12] = xif.to(device)
+            rows_buf[12:] = xjf.to(device)
+            og._ext.smo_update(X, y, grad, rows_buf, yi * dai, yj * daj, gamma)
+        else:
+            _grad_update_cpu(X, y, grad, xif, xjf, yi * dai, yj * daj, gamma)
+
+    # intercept from free vectors (global)
+    myg = -(y.double() * grad)
+    free = (alpha > 1e-12) & (alpha < C - 1e-12)
+    ssum = torch.tensor([float(myg[free].sum()), float(free.sum())], dtype=torch.float64)
+    dist.allreduce_(ssum)
+    if float(ssum[1]) > 0:
+        b = float(ssum[0] / ssum[1])
+    else:
+        up = ((y > 0) & (alpha < C)) | ((y < 0) & (alpha > 0))
+        low = ((y > 0) & (alpha > 0)) | ((y < 0) & (alpha < C))
+        hi = torch.tensor(
+            [float(torch.where(up, myg, torch.tensor(-math.inf, dtype=torch.float64, device=device)).max())]
+        )
+        lo = torch.tensor(
+            [float(torch.where(low, myg, torch.tensor(math.inf, dtype=torch.float64, device=device)).min())]
+        )
+        dist.allreduce_(hi, op=torch.distributed.ReduceOp.MAX if dist.is_initialized() else None)
+        dist.allreduce_(lo, op=torch.distributed.ReduceOp.MIN if dist.is_initialized() else None)
+        b = float((hi[0] + lo[0]) / 2.0)
+    return alpha, b, it

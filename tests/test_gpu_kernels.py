@@ -162,8 +162,9 @@ def test_flow_features_kernel():
     prev = torch.from_numpy(rng.integers(0, 10_000, (n, 4)).astype(np.float64))
     cur = prev + torch.from_numpy(rng.integers(0, 1000, (n, 4)).astype(np.float64))
     t0 = torch.full((n, 1), 100.0, dtype=torch.float64)
-    times = torch.cat([t0 + 10, t0 + 9, t0 + 9, t0], dim=1)
-    # some rows with zero deltas (division guards)
+    # [tf_cur, tf_prev, tr_cur, tr_prev, t_start, pad]
+    times = torch.cat([t0 + 10, t0 + 9, t0 + 10, t0 + 8, t0, t0 * 0], dim=1)
+    # some rows with zero time deltas (division guards)
     times[:100, 1] = times[:100, 0]
     ref = oc.flow_features(cur, prev, times)
     got = og.flow_features(cur.cuda(), prev.cuda(), times.cuda())

@@ -110,7 +110,7 @@ def rf_pack(forest: Dict[str, torch.Tensor], device) -> Dict[str, torch.Tensor]:
     n_trees = offsets.numel() - 1
     base = torch.repeat_interleave(offsets[:-1], offsets[1:] - offsets[:-1])
     is_leaf = feat < 0
-    probs = forest["leaf_proba"]
+    probs = forest["leaf_proba"].cpu()
     # pure leaf: exactly one class carries the whole mass
     pmax, pcls = probs.max(dim=1)
     leaf_pure = torch.zeros_like(is_leaf)

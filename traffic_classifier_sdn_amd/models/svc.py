@@ -134,6 +134,15 @@ class SVC(Estimator):
 
     def predict_index(self, X: ArrayLike) -> torch.Tensor:
         Xt = as_tensor(X, self.device, torch.float32)
+        sc = getattr(self, "_svclass", None)
+        if sc is None or sc.device != Xt.device:
+            # SV -> class byte map; precomputed so the hipGraph-captured
+            # serve path never runs repeat_interleave inside capture
+            sc = torch.repeat_interleave(
+                torch.arange(self.n_support_.numel(), device=Xt.device),
+                self.n_support_.to(Xt.device),
+            ).to(torch.uint8).contiguous()
+            self._svclass = sc
         return ops.svc_predict(
             Xt,
             self.support_vectors_.to(Xt.dtype),
@@ -141,6 +150,7 @@ class SVC(Estimator):
             self.intercept_.to(Xt.dtype),
             self.n_support_,
             self.gamma_,
+            svclass=sc,
         )
 
     def decision_function_ovo(self, X: ArrayLike) -> torch.Tensor:

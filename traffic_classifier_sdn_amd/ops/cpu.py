@@ -216,6 +216,7 @@ def svc_predict(
     intercept: torch.Tensor,
     n_support: torch.Tensor,
     gamma: float,
+    svclass: torch.Tensor = None,
 ) -> torch.Tensor:
     K = rbf_kernel(X, SV, gamma)
     dec = svc_ovo_decision(K, dual_coef, intercept, n_support)

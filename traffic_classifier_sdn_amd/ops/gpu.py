@@ -82,10 +82,14 @@ def svc_predict(
     intercept: torch.Tensor,
     n_support: torch.Tensor,
     gamma: float,
+    svclass: torch.Tensor = None,
 ) -> torch.Tensor:
-    svclass = torch.repeat_interleave(
-        torch.arange(n_support.numel(), device=X.device), n_support.to(X.device)
-    ).to(torch.uint8)
+    if svclass is None:
+        # static per model; callers on the hipGraph path precompute it
+        # (repeat_interleave is not stream-capture safe)
+        svclass = torch.repeat_interleave(
+            torch.arange(n_support.numel(), device=X.device), n_support.to(X.device)
+        ).to(torch.uint8)
     return _ext.svc_predict(
         _f32(X), _f32(SV), _f32(dual_coef), svclass.contiguous(), _f32(intercept), float(gamma)
     )

@@ -1,0 +1,91 @@
+"""Offline training pipeline (fit.py — the notebooks' C9 loop, SURVEY.md
+§3.4) and the SVC / RandomForest sklearn-pickle exporters."""
+
+import os
+import pickle
+import subprocess
+import sys
+import warnings
+
+import numpy as np
+import pytest
+
+from traffic_classifier_sdn_amd import fit as fitmod
+from traffic_classifier_sdn_amd.models import SVC, RandomForestClassifier, from_params
+from traffic_classifier_sdn_amd.utils import checkpoint as ckpt
+from traffic_classifier_sdn_amd.utils.datasets import (
+    load_reference_dataset,
+    train_test_split_ref,
+)
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture(scope="module")
+def split():
+    X, y = load_reference_dataset()
+    return train_test_split_ref(X, y)
+
+
+def test_fit_cli_fast_algos(tmp_path, split):
+    """gaussiannb + randomforest + kmeans via the CLI; accuracy reported and
+    checkpoints written under the reference file names."""
+    out = subprocess.run(
+        [
+            sys.executable,
+            "-m",
+            "traffic_classifier_sdn_amd.fit",
+            "--algos",
+            "gaussiannb,randomforest,kmeans",
+            "--out",
+            str(tmp_path),
+            "--json",
+        ],
+        cwd=REPO,
+        capture_output=True,
+        text=True,
+        timeout=300,
+    )
+    assert out.returncode == 0, out.stderr
+    import json
+
+    results = {json.loads(l)["algo"]: json.loads(l) for l in out.stdout.splitlines() if l.startswith("{")}
+    assert results["gaussiannb"]["accuracy"] > 0.97
+    assert results["randomforest"]["accuracy"] > 0.99
+    assert (tmp_path / "GaussianNB.npz").exists()
+    assert (tmp_path / "RandomForestClassifier.npz").exists()
+    assert (tmp_path / "KMeans_Clustering.npz").exists()
+    # KMeans checkpoint carries the mode-based cluster->class name map
+    p = ckpt.load_params_npz(str(tmp_path / "KMeans_Clustering.npz"))
+    assert "cluster_label_names" in p and len(p["cluster_label_names"]) == 6
+
+
+def test_mode_cluster_assignment():
+    ids = np.array([0, 0, 0, 1, 1, 2])
+    y = np.array(["a", "a", "b", "b", "b", "c"], dtype=object)
+    names = fitmod.mode_cluster_assignment(ids, y, 3)
+    assert list(names) == ["a", "b", "c"]
+
+
+def test_rf_sklearn_export_round_trip(tmp_path, split):
+    Xtr, Xte, ytr, yte = split
+    m = RandomForestClassifier(n_estimators=10, seed=7).fit(Xtr, ytr)
+    path = str(tmp_path / "RandomForestClassifier")
+    ckpt.save_sklearn_pickle(m.to_params(), path)
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        sk = pickle.load(open(path, "rb"))
+        agree = (m.predict(Xte) == sk.predict(Xte)).mean()
+    assert agree == 1.0
+
+
+def test_svc_sklearn_export_round_trip(tmp_path, split):
+    Xtr, Xte, ytr, yte = split
+    m = SVC().fit(Xtr[:800], ytr[:800])
+    path = str(tmp_path / "SVC")
+    ckpt.save_sklearn_pickle(m.to_params(), path)
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        sk = pickle.load(open(path, "rb"))
+        agree = (m.predict(Xte) == sk.predict(Xte)).mean()
+    assert agree > 0.999

@@ -1,0 +1,205 @@
+"""Offline training pipeline — the notebooks' fit/eval/save loop as a
+runnable module (reference: models/notebooks.zip → notebooks/*.ipynb,
+SURVEY.md §2.1 C9 / §3.4).
+
+    python -m traffic_classifier_sdn_amd.fit [--algos all] [--out models]
+        [--device cuda|cpu] [--sklearn-pickles] [--test-size 0.5] [--seed 101]
+
+Per algorithm it reproduces the notebook protocol exactly: load the shipped
+per-class CSVs, dropna, drop the 4 cumulative columns (12 model features),
+`train_test_split(test_size=0.5, random_state=101)`, fit, report held-out
+accuracy + confusion matrix, and write the checkpoint under the reference's
+file name (npz engine format, plus an sklearn-1.0.1-layout pickle with
+--sklearn-pickles so stock sklearn can load the result back).
+
+Runs under torchrun too: with a process group initialized, the row shards
+split across ranks and every estimator fits data-parallel (RCCL on GPU,
+gloo on CPU); rank 0 evaluates and writes checkpoints.
+
+KMeans additionally records the notebook's mode-based cluster→class
+assignment (notebooks/1_log_Kmeans.ipynb cell 125): each cluster takes the
+majority true class of its member rows, giving the supervised-accuracy
+number and a meaningful label map for the serve path (the reference's fixed
+index→name map at traffic_classifier.py:109-114 silently mislabels —
+SURVEY.md §2.1 quirks).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+from typing import Dict, Optional, Sequence
+
+import numpy as np
+
+from .models import (
+    GaussianNB,
+    KMeans,
+    KNeighborsClassifier,
+    LogisticRegression,
+    RandomForestClassifier,
+    SVC,
+)
+from .parallel import dist
+from .utils.datasets import load_reference_dataset, train_test_split_ref
+from .utils.metrics import accuracy, confusion_matrix
+from .utils.schema import CLASS_NAMES
+
+# algo key -> (constructor, reference checkpoint file name)
+ALGOS = {
+    "logistic": (lambda device: LogisticRegression(device=device), "LogisticRegression"),
+    "kmeans": (lambda device: KMeans(n_clusters=len(CLASS_NAMES), device=device), "KMeans_Clustering"),
+    "svm": (lambda device: SVC(device=device), "SVC"),
+    "knearest": (lambda device: KNeighborsClassifier(device=device), "KNeighbors"),
+    "randomforest": (lambda device: RandomForestClassifier(device=device), "RandomForestClassifier"),
+    "gaussiannb": (lambda device: GaussianNB(device=device), "GaussianNB"),
+}
+
+# published notebook accuracies to match-or-beat
+PUBLISHED_ACC = {
+    "logistic": 0.9647,
+    "svm": 0.8501,
+    "randomforest": 0.9987,
+    "knearest": 0.9930,
+    "gaussiannb": 0.9863,
+}
+
+
+def mode_cluster_assignment(cluster_ids: np.ndarray, y_true: np.ndarray, n_clusters: int):
+    """Majority-vote cluster→class map (1_log_Kmeans.ipynb cell 125,
+    scipy.stats.mode over each cluster's true labels)."""
+    classes = np.unique(y_true)
+    cls_index = {c: i for i, c in enumerate(classes)}
+    y_idx = np.asarray([cls_index[v] for v in y_true])
+    label_map = np.zeros(n_clusters, dtype=np.int64)
+    for c in range(n_clusters):
+        members = y_idx[cluster_ids == c]
+        label_map[c] = np.bincount(members, minlength=len(classes)).argmax() if len(members) else 0
+    return classes[label_map]  # cluster id -> class name
+
+
+def fit_one(
+    algo: str,
+    Xtr: np.ndarray,
+    ytr: np.ndarray,
+    Xte: np.ndarray,
+    yte: np.ndarray,
+    device: Optional[str],
+    sharded: bool,
+) -> Dict:
+    ctor, _ = ALGOS[algo]
+    model = ctor(device)
+    rank = dist.rank() if dist.is_initialized() else 0
+    # row shard for this rank (deterministic contiguous split)
+    if sharded:
+        world = dist.world_size()
+        Xl = np.array_split(Xtr, world)[rank]
+        yl = np.array_split(ytr, world)[rank]
+    else:
+        Xl, yl = Xtr, ytr
+    t0 = time.perf_counter()
+    if algo == "kmeans":
+        model.fit(Xl, sharded=sharded)
+    elif algo == "randomforest":
+        model.fit(Xtr, ytr)  # RF fit is tree-parallel internally, full rows
+    else:
+        model.fit(Xl, yl, sharded=sharded)
+    fit_s = time.perf_counter() - t0
+
+    result = {"algo": algo, "fit_seconds": fit_s}
+    if rank == 0:
+        if algo == "kmeans":
+            # supervised evaluation of the unsupervised fit via the
+            # mode-based assignment computed on the TRAIN rows
+            tr_clusters = model.predict_index(Xtr).cpu().numpy()
+            cluster_to_name = mode_cluster_assignment(tr_clusters, ytr, model.n_clusters)
+            model.cluster_label_names_ = cluster_to_name
+            te_clusters = model.predict_index(Xte).cpu().numpy()
+            pred = cluster_to_name[te_clusters]
+        else:
+            pred = model.predict(Xte)
+        acc = accuracy(yte, pred)
+        cm = confusion_matrix(yte, pred, labels=list(CLASS_NAMES))
+        result["accuracy"] = acc
+        result["confusion_matrix"] = cm.tolist()
+        result["published_accuracy"] = PUBLISHED_ACC.get(algo)
+    result["_model"] = model
+    return result
+
+
+def main(argv: Optional[Sequence[str]] = None) -> int:
+    ap = argparse.ArgumentParser(description=__doc__)
+    ap.add_argument("--algos", default="all", help="comma list of %s or 'all'" % ",".join(ALGOS))
+    ap.add_argument("--out", default="models", help="checkpoint output directory")
+    ap.add_argument("--device", default=None, help="cpu / cuda (default: auto)")
+    ap.add_argument("--data-dir", default=None, help="training CSV directory (default: shipped dataset)")
+    ap.add_argument("--test-size", type=float, default=0.5)
+    ap.add_argument("--seed", type=int, default=101)
+    ap.add_argument("--sklearn-pickles", action="store_true", help="also write sklearn-1.0.1-layout pickles")
+    ap.add_argument("--json", action="store_true", help="machine-readable result lines")
+    args = ap.parse_args(argv)
+
+    # torchrun-aware: initialize the group if launched with a WORLD_SIZE
+    sharded = False
+    if int(os.environ.get("WORLD_SIZE", "1")) > 1 and not dist.is_initialized():
+        import torch
+
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_from_env(backend)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(dist.env_local_rank())
+    sharded = dist.is_initialized() and dist.world_size() > 1
+    rank = dist.rank() if dist.is_initialized() else 0
+
+    X, y = load_reference_dataset(data_dir=args.data_dir)
+    Xtr, Xte, ytr, yte = train_test_split_ref(X, y, test_size=args.test_size, random_state=args.seed)
+
+    algos = list(ALGOS) if args.algos == "all" else [a.strip().lower() for a in args.algos.split(",")]
+    for a in algos:
+        if a not in ALGOS:
+            print(f"ERROR: unknown algo {a!r}; choices: {', '.join(ALGOS)}", file=sys.stderr)
+            return 2
+
+    os.makedirs(args.out, exist_ok=True) if rank == 0 else None
+    ok = True
+    for algo in algos:
+        res = fit_one(algo, Xtr, ytr, Xte, yte, args.device, sharded)
+        model = res.pop("_model")
+        if rank == 0:
+            fname = ALGOS[algo][1]
+            npz_path = os.path.join(args.out, fname + ".npz")
+            model.save(npz_path)
+            if args.sklearn_pickles:
+                try:
+                    model.save(os.path.join(args.out, fname))  # no .npz -> sklearn pickle
+                except NotImplementedError as e:
+                    print(f"WARNING: {algo}: {e}", file=sys.stderr)
+            res["checkpoint"] = npz_path
+            pub = res.get("published_accuracy")
+            if pub is not None and res["accuracy"] < pub - 0.02:
+                ok = False
+            if args.json:
+                print(json.dumps({k: v for k, v in res.items() if k != "confusion_matrix"}))
+            else:
+                line = f"{algo:13s} acc={res['accuracy']:.4f}"
+                if pub is not None:
+                    line += f"  (published {pub:.4f})"
+                line += f"  fit={res['fit_seconds']:.2f}s  -> {npz_path}"
+                print(line)
+                cm = np.asarray(res["confusion_matrix"])
+                hdr = " ".join(f"{c[:6]:>7s}" for c in CLASS_NAMES)
+                print(f"{'':13s} confusion (rows=true):\n{'':16s}{hdr}")
+                for cname, row in zip(CLASS_NAMES, cm):
+                    print(f"{'':14s}{cname[:6]:>7s} " + " ".join(f"{v:7d}" for v in row))
+    if dist.is_initialized():
+        import torch
+
+        torch.distributed.destroy_process_group()
+    return 0 if ok else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())

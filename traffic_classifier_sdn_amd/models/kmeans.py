@@ -124,13 +124,17 @@ class KMeans(Estimator):
 
     # -- checkpointing -------------------------------------------------
     def to_params(self) -> Dict[str, Any]:
-        return {
+        p = {
             "kind": self.kind,
             "centers": self.cluster_centers_.double().cpu().numpy(),
             "n_clusters": int(self.n_clusters),
             "inertia": float(self.inertia_),
             "n_iter": int(self.n_iter_),
         }
+        if getattr(self, "cluster_label_names_", None) is not None:
+            # mode-based cluster -> class-name map (fit.py, nb1 cell 125)
+            p["cluster_label_names"] = np.asarray(self.cluster_label_names_, dtype=object)
+        return p
 
     @classmethod
     def from_params(cls, params: Dict[str, Any], device: Optional[str] = None):
@@ -139,4 +143,6 @@ class KMeans(Estimator):
         m.cluster_centers_ = torch.as_tensor(centers).to(m.device)
         m.inertia_ = float(params.get("inertia", 0.0))
         m.n_iter_ = int(params.get("n_iter", 0))
+        if params.get("cluster_label_names") is not None:
+            m.cluster_label_names_ = np.asarray(params["cluster_label_names"], dtype=object)
         return m

@@ -306,12 +306,36 @@ def params_to_sklearn(params: Dict[str, Any]):
         est._n_threads = 1
         return est
     if kind == "svc":
-        # Rebuilding libsvm internals across versions is fragile; export the
-        # decision data via a thin sklearn-API wrapper instead.
-        raise NotImplementedError(
-            "SVC export to sklearn object is not supported; use the "
-            "framework-native .npz checkpoint (save_params_npz)."
-        )
+        from sklearn.svm import SVC as SkSVC
+
+        SV = np.ascontiguousarray(params["support_vectors"], dtype=np.float64)
+        dual = np.ascontiguousarray(params["dual_coef"], dtype=np.float64)
+        intercept = np.ascontiguousarray(params["intercept"], dtype=np.float64)
+        n_support = np.ascontiguousarray(params["n_support"], dtype=np.int32)
+        classes = np.asarray(params["classes"])
+        gamma = float(params["gamma"])
+        est = SkSVC(kernel="rbf", gamma=gamma)
+        est.classes_ = classes
+        est.support_vectors_ = SV
+        support = np.asarray(params.get("support", np.arange(SV.shape[0])))
+        est.support_ = np.ascontiguousarray(support, dtype=np.int32)
+        est._n_support = n_support
+        # sklearn negates the public dual/intercept for binary problems
+        # (sklearn/svm/_base.py fit); params hold the public layout
+        binary = len(classes) == 2
+        est.dual_coef_ = dual
+        est._dual_coef_ = -dual if binary else dual
+        est.intercept_ = intercept
+        est._intercept_ = -intercept if binary else intercept
+        est._probA = np.empty(0, dtype=np.float64)
+        est._probB = np.empty(0, dtype=np.float64)
+        est._gamma = gamma
+        est._sparse = False
+        est.shape_fit_ = (int(params.get("n_fit_rows", SV.shape[0])), SV.shape[1])
+        est.fit_status_ = 0
+        est.n_features_in_ = SV.shape[1]
+        est._num_iter = np.ones(max(1, len(intercept)), dtype=np.int32)
+        return est
     if kind == "kneighbors":
         from sklearn.neighbors import KNeighborsClassifier
 
@@ -319,10 +343,69 @@ def params_to_sklearn(params: Dict[str, Any]):
         est.fit(np.asarray(params["fit_X"], dtype=np.float64), np.asarray(params["classes"])[np.asarray(params["y"])])
         return est
     if kind == "random_forest":
-        raise NotImplementedError(
-            "RandomForest export to sklearn object is not supported; use the "
-            "framework-native .npz checkpoint (save_params_npz)."
-        )
+        from sklearn.ensemble import RandomForestClassifier as SkRF
+        from sklearn.tree import DecisionTreeClassifier
+        from sklearn.tree._tree import Tree
+
+        classes = np.asarray(params["classes"])
+        C = len(classes)
+        trees = params["trees"]
+        F = int(params.get("n_features", max(int(t["feature"].max()) for t in trees) + 1))
+        # version-robust templates: sklearn's node struct dtype and the
+        # count-vs-fraction convention of tree values (changed in 1.4)
+        tmpl = DecisionTreeClassifier(max_depth=1).fit([[0.0], [1.0]], [0, 1])
+        tstate = tmpl.tree_.__getstate__()
+        node_dtype = tstate["nodes"].dtype
+        values_normalized = abs(float(tstate["values"][0].sum()) - 1.0) < 1e-9
+        ests = []
+        for t in trees:
+            feat = np.asarray(t["feature"], dtype=np.int64)
+            n = len(feat)
+            left = np.asarray(t["left"], dtype=np.int64)
+            right = np.asarray(t["right"], dtype=np.int64)
+            thr = np.asarray(t["threshold"], dtype=np.float64)
+            counts = np.asarray(t["values"], dtype=np.float64)
+            nns = counts.sum(axis=1)
+            p = counts / np.maximum(nns[:, None], 1e-300)
+            is_leaf = feat < 0
+            nodes = np.zeros(n, dtype=node_dtype)
+            nodes["left_child"] = np.where(is_leaf, -1, left)
+            nodes["right_child"] = np.where(is_leaf, -1, right)
+            nodes["feature"] = np.where(is_leaf, -2, feat)
+            nodes["threshold"] = np.where(is_leaf, -2.0, thr)
+            nodes["impurity"] = 1.0 - (p ** 2).sum(axis=1)
+            nodes["n_node_samples"] = nns.astype(np.int64)
+            nodes["weighted_n_node_samples"] = nns
+            # depth per node via parent scan (children follow parents)
+            depth = np.zeros(n, dtype=np.int64)
+            for i in range(n):
+                if feat[i] >= 0:
+                    depth[left[i]] = depth[i] + 1
+                    depth[right[i]] = depth[i] + 1
+            tree = Tree(F, np.asarray([C], dtype=np.intp), 1)
+            tree.__setstate__(
+                {
+                    "max_depth": int(depth.max()),
+                    "node_count": n,
+                    "nodes": nodes,
+                    "values": (p if values_normalized else counts).reshape(n, 1, C),
+                }
+            )
+            dt = DecisionTreeClassifier()
+            dt.tree_ = tree
+            dt.classes_ = classes
+            dt.n_classes_ = C
+            dt.n_features_in_ = F
+            dt.n_outputs_ = 1
+            dt.max_features_ = max(1, int(np.sqrt(F)))
+            ests.append(dt)
+        est = SkRF(n_estimators=len(trees))
+        est.estimators_ = ests
+        est.classes_ = classes
+        est.n_classes_ = C
+        est.n_features_in_ = F
+        est.n_outputs_ = 1
+        return est
     raise ValueError(f"unknown params kind {kind}")
 
 

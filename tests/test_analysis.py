@@ -1,0 +1,59 @@
+"""PCA / cluster analysis module (C10, notebooks/1_log_Kmeans.ipynb cells
+63-131) — numerics validated against sklearn's PCA on the same rows."""
+
+import numpy as np
+import pytest
+import torch
+
+from traffic_classifier_sdn_amd import analysis
+from traffic_classifier_sdn_amd.utils.datasets import load_reference_dataset
+
+
+@pytest.fixture(scope="module")
+def Xs():
+    X, _ = load_reference_dataset()
+    Xt = torch.as_tensor(X, dtype=torch.float64)
+    Xs, _, _ = analysis.standardize(Xt)
+    return Xs
+
+
+def test_standardize_matches_sklearn(Xs):
+    sk = pytest.importorskip("sklearn.preprocessing")
+    X, _ = load_reference_dataset()
+    ref = sk.StandardScaler().fit_transform(X)
+    np.testing.assert_allclose(Xs.numpy(), ref, atol=1e-9)
+
+
+def test_pca_matches_sklearn(Xs):
+    skd = pytest.importorskip("sklearn.decomposition")
+    comps, ratio, proj = analysis.pca(Xs, n_components=2)
+    ref = skd.PCA(n_components=2).fit(Xs.numpy())
+    np.testing.assert_allclose(ratio.numpy(), ref.explained_variance_ratio_, rtol=1e-9)
+    # components match up to sign
+    for i in range(2):
+        a, b = comps[i].numpy(), ref.components_[i]
+        assert min(np.abs(a - b).max(), np.abs(a + b).max()) < 1e-8
+    refproj = ref.transform(Xs.numpy())
+    for i in range(2):
+        a, b = proj[:, i].numpy(), refproj[:, i]
+        assert min(np.abs(a - b).max(), np.abs(a + b).max()) < 1e-6
+
+
+def test_run_analysis_end_to_end():
+    res = analysis.run_analysis(device="cpu")
+    assert 0.7 < res["pca_explained_variance_total"] < 0.95
+    assert res["lr_accuracy_on_2pc"] > 0.8  # notebook: 0.8303
+    assert len(res["kmeans_mode_assignment"]) == 6
+    assert res["kmeans_supervised_accuracy"] > 0.4
+
+
+def test_ditg_script_generation(tmp_path):
+    from traffic_classifier_sdn_amd import ditg
+
+    paths = ditg.write_scripts(str(tmp_path))
+    names = sorted(p.split("/")[-1] for p in paths)
+    assert "all_script_file" in names and "voice_script_file" in names
+    all_lines = open(tmp_path / "all_script_file").read().splitlines()
+    assert len(all_lines) == 5
+    assert all_lines[0] == "-a 10.0.0.1 -rp 10001 VoIP -x G.711.2 -h RTP -VAD"
+    assert open(tmp_path / "quake_script_file").read().strip() == "-a 10.0.0.1 -rp 10002 Quake3"

@@ -1,19 +1,26 @@
-"""Flagship benchmark: RandomForest (100 trees, 6-class) flow classification
-throughput — flows/sec on synthetic flow-stat rows (BASELINE.json config #2),
-weak-scaled over N GPUs (one rank per GPU, RCCL).
+"""Benchmarks for the BASELINE.json configs on MI355X.
 
+Flagship (driver contract, default):
     python bench.py --gpus N --steps K --warmup W [--rows-per-gpu R]
+runs the RandomForest (100 trees, 6-class) predict throughput on synthetic
+flow-stat rows (config #2), weak-scaled over N GPUs (one rank per GPU, RCCL).
+Rank 0 prints ONE JSON line; elapsed is the MAX over ranks, timed region
+bracketed by barrier + torch.cuda.synchronize on both sides.
 
-Per step, every rank classifies its resident R-row shard with the packed-
-forest HIP traversal kernel (ops.gpu.rf_argmax).  The timed region is
-bracketed by a barrier + torch.cuda.synchronize on both sides; elapsed time
-is the MAX over ranks; rank 0 prints one JSON line.
+Extra workloads (same JSON contract, run explicitly):
+    --workload knn       config #4: k=5 brute-force, reference set sharded
+                         across ranks (12.5M rows/GPU -> 100M at 8), per-step
+                         top-k + RCCL all-gather merge over 64K query rows
+    --workload svc-fit   config #3: RBF-SVC SMO on 1M rows strong-scaled
+                         across ranks; per step a fixed block of SMO
+                         iterations (fused kernel-row gradient updates)
+    --workload serve     config #5: full poll cycle — counter snapshot ->
+                         H2D -> hipGraph replay of feature-extract + 5-model
+                         ensemble predict -> D2H labels, 8192 live flows
 
-The forest is the reference's own 100-tree 6-class checkpoint (converted to
-data/ref_models/RandomForestClassifier.npz), so tree shapes/depths match the
-named config; rows are drawn from the real flow-stat rows with jitter.  The
-model's 6-class accuracy on the real held-out split is reported alongside
-(99.87%).
+The forest/model shapes are the reference's own checkpoints (converted to
+data/ref_models/*.npz), so model config matches the named baseline; data is
+synthetic (no network for datasets) with the real rows' scale/jitter.
 """
 
 from __future__ import annotations
@@ -42,12 +49,243 @@ REPO = os.path.dirname(os.path.abspath(__file__))
 RF_CKPT = os.path.join(REPO, "data", "ref_models", "RandomForestClassifier.npz")
 
 
+def _timed(step, steps: int, warmup: int, use_gpu: bool) -> float:
+    """Warmup, barrier+sync, time K steps, barrier+sync; MAX over ranks."""
+    for _ in range(warmup):
+        step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    dist.barrier()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    e = torch.tensor([elapsed], dtype=torch.float64)
+    if dist.is_initialized():
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+    dist.barrier()
+    return float(e[0])
+
+
+def _emit(rank, metric, value, unit, world, args, ms_per_step, scaling, config):
+    if rank != 0:
+        return
+    print(
+        json.dumps(
+            {
+                "metric": metric,
+                "value": value,
+                "unit": unit,
+                "n_gpus": world,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": ms_per_step,
+                "higher_is_better": True,
+                "scaling": scaling,
+                "vs_baseline": None,
+                "dtype": "fp32",
+                "data": "synthetic",
+                "config": config,
+            }
+        )
+    )
+
+
+def bench_rf(args, rank, world, device, use_gpu):
+    model = load_model(RF_CKPT, device=device)
+    # measured (non-timed) accuracy on the real data, reference split
+    X_real, y_real = load_reference_dataset()
+    _, Xte, _, yte = train_test_split_ref(X_real, y_real)
+    acc = accuracy(yte, model.predict(Xte))
+
+    Xn = synthetic_flow_rows(args.rows_per_gpu, seed=args.seed + 1000 * rank, reference_X=X_real)
+    X = torch.from_numpy(Xn).to(device)
+    if use_gpu:
+        torch.cuda.synchronize()
+
+    elapsed = _timed(lambda: model.predict_index(X), args.steps, args.warmup, use_gpu)
+    total_rows = args.rows_per_gpu * world * args.steps
+    _emit(
+        rank,
+        "flows/sec, RandomForest 100-tree 6-class predict on synthetic flow-stat rows",
+        total_rows / elapsed,
+        "flows/s",
+        world,
+        args,
+        elapsed / args.steps * 1000.0,
+        "weak",
+        {
+            "model": "RandomForestClassifier-100trees-6class",
+            "global_batch": args.rows_per_gpu * world,
+            "seq_len": 12,
+            "parallelism": f"dp{world}",
+            "accuracy_6class": acc,
+            "accuracy_published_ref": 0.9987,
+        },
+    )
+
+
+def bench_knn(args, rank, world, device, use_gpu):
+    """Config #4: sharded-reference brute-force KNN with all-gather merge."""
+    from traffic_classifier_sdn_amd.models import KNeighborsClassifier
+
+    ref_rows = args.knn_ref_rows_per_gpu
+    X_real, _ = load_reference_dataset()
+    Xref = synthetic_flow_rows(ref_rows, seed=args.seed + 7000 + 1000 * rank, reference_X=X_real)
+    yref = np.random.default_rng(args.seed + rank).integers(0, 6, size=ref_rows)
+    m = KNeighborsClassifier(n_neighbors=5, batch_rows=args.knn_queries, device=device)
+    m.fit(Xref, yref, sharded=world > 1)
+    Q = torch.from_numpy(
+        synthetic_flow_rows(args.knn_queries, seed=args.seed + 31 + rank, reference_X=X_real)
+    ).to(device)
+    if use_gpu:
+        torch.cuda.synchronize()
+
+    elapsed = _timed(lambda: m.predict_index(Q), args.steps, args.warmup, use_gpu)
+    total_q = args.knn_queries * world * args.steps
+    _emit(
+        rank,
+        "queries/sec, KNeighbors k=5 brute-force over sharded reference set (RCCL all-gather top-k)",
+        total_q / elapsed,
+        "queries/s",
+        world,
+        args,
+        elapsed / args.steps * 1000.0,
+        "weak",
+        {
+            "model": "KNeighbors-k5",
+            "global_batch": args.knn_queries * world,
+            "seq_len": 12,
+            "parallelism": f"shard{world}",
+            "reference_rows_total": ref_rows * world,
+            "reference_rows_per_gpu": ref_rows,
+        },
+    )
+
+
+def bench_svc_fit(args, rank, world, device, use_gpu):
+    """Config #3: SMO iteration throughput on 1M rows, strong-scaled."""
+    from traffic_classifier_sdn_amd.models.svc_fit import smo_fit_pair
+
+    n_total = args.svc_rows
+    n_local = n_total // world
+    rng = np.random.default_rng(args.seed + rank)
+    X_real, _ = load_reference_dataset()
+    Xn = synthetic_flow_rows(n_local, seed=args.seed + 50 + rank, reference_X=X_real).astype(np.float32)
+    # separable-ish binary labels over two features (one OVO subproblem)
+    y = np.where(Xn[:, 0] + 0.3 * Xn[:, 6] > np.median(Xn[:, 0]), 1.0, -1.0).astype(np.float32)
+    X = torch.from_numpy(Xn).to(device)
+    yt = torch.from_numpy(y).to(device)
+    gamma = 1.0 / (12 * float(X.var()))
+    iters = args.svc_iters_per_step
+    if use_gpu:
+        torch.cuda.synchronize()
+
+    done = []
+
+    def step():
+        _, _, it = smo_fit_pair(X, yt, C=1.0, gamma=gamma, tol=0.0, max_iter=iters)
+        done.append(it)
+
+    elapsed = _timed(step, args.steps, args.warmup, use_gpu)
+    timed_iters = sum(done[args.warmup :])
+    # each SMO iteration updates the full gradient: n_total fused kernel-row
+    # evaluations -> row-updates/s is the scale-invariant rate
+    row_updates = float(timed_iters) * n_total / max(1, world) * world
+    _emit(
+        rank,
+        "SMO row-updates/sec, RBF-SVC dual fit on 1M synthetic flow rows (fused kernel-row gradient)",
+        row_updates / elapsed,
+        "row-updates/s",
+        world,
+        args,
+        elapsed / args.steps * 1000.0,
+        "strong",
+        {
+            "model": "RBF-SVC-SMO",
+            "global_batch": n_total,
+            "seq_len": 12,
+            "parallelism": f"dp{world}",
+            "rows_total": n_total,
+            "smo_iters_per_step": iters,
+        },
+    )
+
+
+def bench_serve(args, rank, world, device, use_gpu):
+    """Config #5: hipGraph-captured poll cycle over 8192 live flows."""
+    from traffic_classifier_sdn_amd.flow.parser import replay
+    from traffic_classifier_sdn_amd.flow.replay import SynthFlowSpec, TelemetryReplaySource
+    from traffic_classifier_sdn_amd.serve_gpu import GpuServeEngine
+
+    n_flows = args.serve_flows
+    rng = np.random.default_rng(args.seed)
+    specs = [
+        SynthFlowSpec(
+            "02:%02x:%02x:%02x:%02x:%02x" % tuple(int(v) for v in rng.integers(0, 256, 5)),
+            "06:%02x:%02x:%02x:%02x:%02x" % tuple(int(v) for v in rng.integers(0, 256, 5)),
+            float(rng.uniform(1, 60)),
+            float(rng.uniform(60, 1200)),
+            float(rng.uniform(1, 60)),
+            float(rng.uniform(60, 1200)),
+        )
+        for _ in range(n_flows)
+    ]
+    src = TelemetryReplaySource(specs=specs, seed=args.seed)
+    table = replay(src.stream(2))
+    names = ["RandomForestClassifier", "GaussianNB", "LogisticRegression", "SVC", "KMeans_Clustering"]
+    models = {
+        n: load_model(os.path.join(REPO, "data", "ref_models", n + ".npz"), device=device)
+        for n in names
+    }
+    eng = GpuServeEngine(models, capacity=n_flows, use_graph=use_gpu, device=device)
+
+    lat = []
+
+    def step():
+        eng.classify(table)
+        lat.append(eng.last_latency_s)
+
+    elapsed = _timed(step, args.steps, args.warmup, use_gpu)
+    lat_t = np.asarray(lat[args.warmup :])
+    _emit(
+        rank,
+        "flows/sec through full poll cycle (snapshot->hipGraph ensemble predict->labels), 5-model ensemble",
+        n_flows * args.steps * world / elapsed,
+        "flows/s",
+        world,
+        args,
+        elapsed / args.steps * 1000.0,
+        "weak",
+        {
+            "model": "ensemble-5(RF+GNB+LR+SVC+KMeans)",
+            "global_batch": n_flows,
+            "seq_len": 12,
+            "parallelism": f"dp{world}",
+            "poll_latency_ms_p50": float(np.percentile(lat_t, 50) * 1e3),
+            "poll_latency_ms_p99": float(np.percentile(lat_t, 99) * 1e3),
+            "hipgraph": use_gpu,
+        },
+    )
+
+
+WORKLOADS = {"rf": bench_rf, "knn": bench_knn, "svc-fit": bench_svc_fit, "serve": bench_serve}
+
+
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--workload", choices=sorted(WORKLOADS), default="rf")
     ap.add_argument("--rows-per-gpu", type=int, default=10_000_000)
+    ap.add_argument("--knn-ref-rows-per-gpu", type=int, default=12_500_000)
+    ap.add_argument("--knn-queries", type=int, default=65_536)
+    ap.add_argument("--svc-rows", type=int, default=1_000_000)
+    ap.add_argument("--svc-iters-per-step", type=int, default=200)
+    ap.add_argument("--serve-flows", type=int, default=8192)
     ap.add_argument("--seed", type=int, default=0)
     args = ap.parse_args()
 
@@ -59,73 +297,12 @@ def main() -> int:
     else:
         device = "cpu"
         args.rows_per_gpu = min(args.rows_per_gpu, 20_000)
+        args.knn_ref_rows_per_gpu = min(args.knn_ref_rows_per_gpu, 20_000)
+        args.svc_rows = min(args.svc_rows, 20_000)
+        args.serve_flows = min(args.serve_flows, 512)
 
-    model = load_model(RF_CKPT, device=device)
+    WORKLOADS[args.workload](args, rank, world, device, use_gpu)
 
-    # measured (non-timed) accuracy on the real data, reference split
-    X_real, y_real = load_reference_dataset()
-    _, Xte, _, yte = train_test_split_ref(X_real, y_real)
-    acc = accuracy(yte, model.predict(Xte))
-
-    # resident synthetic shard, seeded per rank
-    Xn = synthetic_flow_rows(
-        args.rows_per_gpu, seed=args.seed + 1000 * rank, reference_X=X_real
-    )
-    X = torch.from_numpy(Xn).to(device)
-    if use_gpu:
-        torch.cuda.synchronize()
-
-    def step():
-        out = model.predict_index(X)
-        return out
-
-    for _ in range(args.warmup):
-        step()
-    if use_gpu:
-        torch.cuda.synchronize()
-    dist.barrier()
-    t0 = time.perf_counter()
-    for _ in range(args.steps):
-        step()
-    if use_gpu:
-        torch.cuda.synchronize()
-    elapsed = time.perf_counter() - t0
-    # MAX over ranks
-    e = torch.tensor([elapsed], dtype=torch.float64, device=device if use_gpu else "cpu")
-    if dist.is_initialized():
-        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
-    dist.barrier()
-    elapsed = float(e[0])
-
-    total_rows = args.rows_per_gpu * world * args.steps
-    flows_per_sec = total_rows / elapsed
-    if rank == 0:
-        print(
-            json.dumps(
-                {
-                    "metric": "flows/sec, RandomForest 100-tree 6-class predict on synthetic flow-stat rows",
-                    "value": flows_per_sec,
-                    "unit": "flows/s",
-                    "n_gpus": world,
-                    "steps": args.steps,
-                    "warmup": args.warmup,
-                    "ms_per_step": elapsed / args.steps * 1000.0,
-                    "higher_is_better": True,
-                    "scaling": "weak",
-                    "vs_baseline": None,
-                    "dtype": "fp32",
-                    "data": "synthetic",
-                    "config": {
-                        "model": "RandomForestClassifier-100trees-6class",
-                        "global_batch": args.rows_per_gpu * world,
-                        "seq_len": 12,
-                        "parallelism": f"dp{world}",
-                        "accuracy_6class": acc,
-                        "accuracy_published_ref": 0.9987,
-                    },
-                }
-            )
-        )
     if dist.is_initialized():
         torch.distributed.destroy_process_group()
     return 0

@@ -29,6 +29,7 @@ setup(
             sources=[
                 os.path.join(CSRC, "ext.cpp"),
                 os.path.join(CSRC, "predict_kernels.hip"),
+                os.path.join(CSRC, "knn_mfma.hip"),
                 os.path.join(CSRC, "fit_kernels.hip"),
             ],
             extra_compile_args={

@@ -221,3 +221,39 @@ def test_svc_gpu_fit_accuracy(X_real):
     m = SVC(device="cuda").fit(Xtr, ytr)
     acc = accuracy(yte, m.predict(Xte))
     assert acc > 0.80  # published RBF-SVC: 85.01% (6-class; 5 classes here)
+
+
+@pytest.mark.gpu
+def test_knn_mfma_matches_scalar_kernel():
+    """MFMA distance-GEMM path (csrc/knn_mfma.hip) vs the scalar kernel on a
+    reference set large enough to engage sharding; distances must agree to
+    f32 roundoff and selected neighbours must lie inside the scalar k-th
+    distance boundary (ties between duplicate rows may reorder)."""
+    from traffic_classifier_sdn_amd.ops.gpu import _ext, _knn_cmean
+    from traffic_classifier_sdn_amd.utils.datasets import synthetic_flow_rows
+
+    Rn = synthetic_flow_rows(400_000, seed=3)
+    Qn = synthetic_flow_rows(4096, seed=4)
+    R = torch.from_numpy(Rn).float().cuda().contiguous()
+    Q = torch.from_numpy(Qn).float().cuda().contiguous()
+    k = 5
+    d1, i1 = _ext.knn_topk(Q, R, None, k, 0, 0)
+    d2, i2 = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), None, k, 0, 0, 8)
+    d1, i1, d2, i2 = d1.cpu(), i1.cpu().long(), d2.cpu(), i2.cpu().long()
+    # refined distances are the exact diff-form values: k-th boundary parity
+    bound = d1[:, -1] * (1 + 1e-4) + 1e-3
+    assert bool((d2 <= bound.unsqueeze(1)).all())
+    # and vice versa (the MFMA path found nothing worse than the scalar one)
+    bound2 = d2[:, -1] * (1 + 1e-4) + 1e-3
+    assert bool((d1 <= bound2.unsqueeze(1)).all())
+    # index agreement away from ties
+    agree_idx = (i1 == i2).float().mean().item()
+    assert agree_idx > 0.99, agree_idx
+
+    # fused vote + idx_base (sharded merge contract)
+    y = torch.from_numpy(np.random.default_rng(0).integers(0, 6, size=400_000)).cuda()
+    y8 = y.to(torch.uint8).contiguous()
+    _, im, labm = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), y8, k, 6, 1000, 8)
+    _, is_, labs = _ext.knn_topk(Q, R, y8, k, 6, 1000)
+    assert bool((im.min() >= 1000).item())
+    assert (labm.cpu() == labs.cpu()).float().mean().item() > 0.99

@@ -35,6 +35,10 @@ void launch_svc_predict(const float*, const float*, const float*,
 void launch_knn_topk(const float*, const float*, const unsigned char*, float*,
                      int*, int*, long long, long long, int, int, long long,
                      hipStream_t);
+void launch_knn_mfma(const float*, const float*, const float*,
+                     const unsigned char*, float*, int*, float*, int*, int*,
+                     long long, long long, int, int, int, long long,
+                     hipStream_t);
 void launch_gnb_fit_stats(const double*, const long long*, double*, double*,
                           double*, long long, int, hipStream_t);
 void launch_logistic_grad(const double*, const long long*, const double*,
@@ -168,6 +172,48 @@ static std::vector<torch::Tensor> knn_topk(torch::Tensor Q, torch::Tensor R,
   return {dist, idx};
 }
 
+// MFMA distance-GEMM path: R sharded across gridDim.y, per-shard partial
+// top-k lists merged by a second kernel (knn_mfma.hip).
+static std::vector<torch::Tensor> knn_topk_mfma(torch::Tensor Q,
+                                                torch::Tensor R,
+                                                torch::Tensor cmean,
+                                                c10::optional<torch::Tensor> ry,
+                                                int64_t k, int64_t C,
+                                                int64_t idx_base,
+                                                int64_t n_shards) {
+  CHECK_IN(Q, torch::kFloat32);
+  CHECK_IN(R, torch::kFloat32);
+  CHECK_IN(cmean, torch::kFloat32);
+  TORCH_CHECK(Q.size(1) == 12 && R.size(1) == 12, "rows must be (n,12)");
+  TORCH_CHECK(k >= 1 && k <= 8, "MFMA path supports k <= 8");
+  TORCH_CHECK(cmean.numel() == 12, "cmean must be (12,)");
+  const long long nq = Q.size(0);
+  const long long nr = R.size(0);
+  int S = (int)n_shards;
+  if (S < 1) S = 1;
+  if ((long long)S > (nr + 127) / 128) S = (int)((nr + 127) / 128);
+  auto part_d = torch::empty({S, nq, k}, Q.options());
+  auto part_i = torch::empty({S, nq, k}, Q.options().dtype(torch::kInt32));
+  auto dist = torch::empty({nq, k}, Q.options());
+  auto idx = torch::empty({nq, k}, Q.options().dtype(torch::kInt32));
+  const unsigned char* ry_ptr = nullptr;
+  torch::Tensor lab;
+  int* lab_ptr = nullptr;
+  if (ry.has_value()) {
+    CHECK_IN(ry.value(), torch::kUInt8);
+    ry_ptr = ry.value().data_ptr<unsigned char>();
+    lab = torch::empty({nq}, Q.options().dtype(torch::kInt32));
+    lab_ptr = lab.data_ptr<int>();
+  }
+  launch_knn_mfma(Q.data_ptr<float>(), R.data_ptr<float>(),
+                  cmean.data_ptr<float>(), ry_ptr, part_d.data_ptr<float>(),
+                  part_i.data_ptr<int>(), dist.data_ptr<float>(),
+                  idx.data_ptr<int>(), lab_ptr, nq, nr, S, (int)k, (int)C,
+                  idx_base, cur_stream());
+  if (ry.has_value()) return {dist, idx, lab};
+  return {dist, idx};
+}
+
 static std::vector<torch::Tensor> gnb_fit_stats(torch::Tensor X,
                                                 torch::Tensor y, int64_t C) {
   CHECK_IN(X, torch::kFloat64);
@@ -248,6 +294,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_predict", &rf_predict, "packed-forest traversal + vote");
   m.def("svc_predict", &svc_predict, "RBF Gram + OVO vote");
   m.def("knn_topk", &knn_topk, "brute-force top-k (+fused vote)");
+  m.def("knn_topk_mfma", &knn_topk_mfma, "MFMA distance-GEMM top-k (+fused vote)");
   m.def("gnb_fit_stats", &gnb_fit_stats, "per-class sufficient stats");
   m.def("logistic_grad", &logistic_grad, "fused CE loss+grad");
   m.def("flow_features", &flow_features, "counters -> 12 features");

@@ -479,30 +479,48 @@ __global__ void knn_topk_kernel(const float* __restrict__ Q,
         for (int i = threadIdx.x; i < cnt; i += blockDim.x) s_y[i] = ry[tile + i];
       __syncthreads();
       if (q >= nq) continue;
-      for (int s = 0; s < cnt; ++s) {
-        float d = 0.f;
+      // branchless sorted insert: slot k takes bd[k-1] when the new element
+      // lands above it, or the new element when it lands here; strict
+      // compares keep earlier (lower) reference indices on ties.
+      auto insert = [&](float d, int ins, int lab) {
 #pragma unroll
-        for (int j = 0; j < F; ++j) {
-          float t = x.v[j] - s_r[s * F + j];
-          d = fmaf(t, t, d);
-        }
-        if (d < bd[K - 1]) {
-          // branchless sorted insert: slot k takes bd[k-1] when the new
-          // element lands above it, or the new element when it lands here;
-          // strict compares keep earlier (lower) reference indices on ties.
-          int ins = (int)(tile + s);
-          int lab = ry ? (int)s_y[s] : 0;
-#pragma unroll
-          for (int k = K - 1; k > 0; --k) {
-            bool above = bd[k - 1] > d;  // new element goes before slot k-1
-            if (bd[k] > d) {
-              bd[k] = above ? bd[k - 1] : d;
-              bi_[k] = above ? bi_[k - 1] : ins;
-              bl[k] = above ? bl[k - 1] : lab;
-            }
+        for (int k = K - 1; k > 0; --k) {
+          bool above = bd[k - 1] > d;  // new element goes before slot k-1
+          if (bd[k] > d) {
+            bd[k] = above ? bd[k - 1] : d;
+            bi_[k] = above ? bi_[k - 1] : ins;
+            bl[k] = above ? bl[k - 1] : lab;
           }
-          if (bd[0] > d) { bd[0] = d; bi_[0] = ins; bl[0] = lab; }
         }
+        if (bd[0] > d) { bd[0] = d; bi_[0] = ins; bl[0] = lab; }
+      };
+      // two candidates per iteration, two split accumulators each: four
+      // independent FMA chains in flight hide VALU and LDS latency (a
+      // single fmaf chain leaves the SIMD idle between dependent ops)
+      auto dist2 = [&](const float* __restrict__ r) {
+        float a = 0.f, b = 0.f;
+#pragma unroll
+        for (int j = 0; j < 6; ++j) {
+          float t = x.v[j] - r[j];
+          a = fmaf(t, t, a);
+        }
+#pragma unroll
+        for (int j = 6; j < F; ++j) {
+          float t = x.v[j] - r[j];
+          b = fmaf(t, t, b);
+        }
+        return a + b;
+      };
+      int s = 0;
+      for (; s + 2 <= cnt; s += 2) {
+        float d0 = dist2(&s_r[s * F]);
+        float d1 = dist2(&s_r[(s + 1) * F]);
+        if (d0 < bd[K - 1]) insert(d0, (int)(tile + s), ry ? (int)s_y[s] : 0);
+        if (d1 < bd[K - 1]) insert(d1, (int)(tile + s + 1), ry ? (int)s_y[s + 1] : 0);
+      }
+      if (s < cnt) {
+        float d0 = dist2(&s_r[s * F]);
+        if (d0 < bd[K - 1]) insert(d0, (int)(tile + s), ry ? (int)s_y[s] : 0);
       }
     }
     if (q >= nq) continue;

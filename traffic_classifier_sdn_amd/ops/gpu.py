@@ -61,8 +61,38 @@ def kmeans_assign(
     return labels, counts.to(dt), sums.to(dt), inertia[0].to(dt)
 
 
+# MFMA path cutover: below this reference-set size the scalar kernel's
+# launch simplicity wins; above it the v_mfma_f32_32x32x2_f32 distance
+# GEMM dominates (csrc/knn_mfma.hip)
+_KNN_MFMA_MIN_ROWS = 100_000
+_knn_cmean_cache: Dict[Tuple[int, int], torch.Tensor] = {}
+
+
+def _knn_cmean(R: torch.Tensor) -> torch.Tensor:
+    """Reference column means (centering makes the expanded-form distance
+    cancellation-safe); cached per reference buffer."""
+    key = (R.data_ptr(), R.shape[0])
+    c = _knn_cmean_cache.get(key)
+    if c is None:
+        if len(_knn_cmean_cache) > 16:
+            _knn_cmean_cache.clear()
+        c = R.mean(dim=0).float().contiguous()
+        _knn_cmean_cache[key] = c
+    return c
+
+
+def _knn_shards(nr: int) -> int:
+    # enough (query-block x shard) workgroups to fill 256 CUs at 2 WG/CU,
+    # with shards of at least ~256K rows to amortize seeding
+    return max(1, min(64, nr // 262_144))
+
+
 def knn_topk(Q: torch.Tensor, R: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor]:
-    dist, idx = _ext.knn_topk(_f32(Q), _f32(R), None, k, 0, 0)
+    Qf, Rf = _f32(Q), _f32(R)
+    if R.shape[0] >= _KNN_MFMA_MIN_ROWS and k <= 8:
+        dist, idx = _ext.knn_topk_mfma(Qf, Rf, _knn_cmean(Rf), None, k, 0, 0, _knn_shards(R.shape[0]))
+        return dist, idx
+    dist, idx = _ext.knn_topk(Qf, Rf, None, k, 0, 0)
     return dist, idx
 
 
@@ -71,7 +101,13 @@ def knn_classify(
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Fused top-k + uniform vote (labels also returned for sharded merge)."""
     y8 = y.to(torch.uint8).contiguous()
-    dist, idx, lab = _ext.knn_topk(_f32(Q), _f32(R), y8, k, n_classes, idx_base)
+    Qf, Rf = _f32(Q), _f32(R)
+    if R.shape[0] >= _KNN_MFMA_MIN_ROWS and k <= 8 and n_classes <= 16:
+        dist, idx, lab = _ext.knn_topk_mfma(
+            Qf, Rf, _knn_cmean(Rf), y8, k, n_classes, idx_base, _knn_shards(R.shape[0])
+        )
+        return dist, idx, lab
+    dist, idx, lab = _ext.knn_topk(Qf, Rf, y8, k, n_classes, idx_base)
     return dist, idx, lab
 
 

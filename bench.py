@@ -31,6 +31,12 @@ import os
 import sys
 import time
 
+# Cap the OpenMP/BLAS pool BEFORE torch import: GPU boxes run under a cgroup
+# CPU quota (cpu.max 16/100ms here) and a full-width spin-waiting thread pool
+# exhausts it, freezing the whole process ~90 ms per period — measured as a
+# p99 latency cliff on the 1 ms serve path (profiles/serve_latency_r01.md).
+os.environ.setdefault("OMP_NUM_THREADS", "4")
+
 import numpy as np
 import torch
 

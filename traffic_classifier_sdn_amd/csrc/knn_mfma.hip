@@ -20,9 +20,9 @@
 // folds the S partial lists per query (+ fused uniform vote).
 //
 // Per workgroup (256 threads = 4 waves):
-//   QB=384 queries staged centered+transposed in LDS (18 KB)
+//   QB=256 queries staged centered+transposed in LDS (12 KB)
 //   R streamed in 128-candidate tiles, centered+transposed (6 KB) + norms
-//   wave w owns query tiles w, w+4, w+8 (32 rows each; row r of a tile is
+//   wave w owns query tiles w, w+4 (32 rows each; row r of a tile is
 //   owned by lanes r and r+32, scanning key columns 0-15 / 16-31)
 
 #include <hip/hip_runtime.h>
@@ -31,7 +31,7 @@
 
 #include "common.h"
 
-#define KM_QB 384      // queries per workgroup (12 query tiles, 3 per wave)
+#define KM_QB 256      // queries per workgroup (8 query tiles, 2 per wave)
 #define KM_TB 128      // candidate tile (4 MFMA column-subtiles)
 #define KM_KMAX 8      // max k supported by this path
 #define KM_F 12
@@ -68,7 +68,10 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
   __shared__ float s_qt[KM_F][KM_QB];            // centered Q^T
   __shared__ float s_rt[KM_F][KM_TB];            // centered R-tile^T
   __shared__ float s_rn[KM_TB];                  // ||r-c||^2 (FLT_MAX pad)
-  __shared__ float s_km[4][32 * KM_PITCH];       // per-wave key matrix [col][row]
+  // per-wave DOUBLE-BUFFERED key matrix [col][row]: the owner scan of
+  // subtile ct-1 overlaps the MFMA + key spill of subtile ct, so the LDS
+  // write->read dependency never stalls the wave
+  __shared__ float s_km[4][2][32 * KM_PITCH];
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -106,11 +109,11 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
 
   // per-(lane, query-tile) top-k sub-list in registers; the lane pair
   // (r, r+32) covers key columns [0,16) / [16,32) of row r
-  float lk[3][KM_KMAX];
-  int li[3][KM_KMAX];
-  float wkey[3];
+  float lk[2][KM_KMAX];
+  int li[2][KM_KMAX];
+  float wkey[2];
 #pragma unroll
-  for (int t = 0; t < 3; ++t) {
+  for (int t = 0; t < 2; ++t) {
 #pragma unroll
     for (int j = 0; j < KM_KMAX; ++j) {
       lk[t][j] = FLT_MAX;
@@ -143,7 +146,7 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
     __syncthreads();
 
 #pragma unroll
-    for (int qti = 0; qti < 3; ++qti) {
+    for (int qti = 0; qti < 2; ++qti) {
       const int qt = wave + 4 * qti;
       const int rowbase = qt * 32;
       // A fragments: lane l -> Qc[rowbase + (l&31)][2s + (l>>5)]
@@ -151,46 +154,52 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
 #pragma unroll
       for (int s = 0; s < 6; ++s) afrag[s] = s_qt[2 * s + half][rowbase + l31];
 
-      for (int ct = 0; ct < KM_TB / 32; ++ct) {
-        f32x16 acc = {};
+      // software pipeline: iteration ct spills subtile ct's keys into
+      // buffer ct&1 and scans subtile ct-1 from the other buffer
+      for (int ct = 0; ct <= KM_TB / 32; ++ct) {
+        if (ct < KM_TB / 32) {
+          f32x16 acc = {};
 #pragma unroll
-        for (int s = 0; s < 6; ++s)
-          acc = __builtin_amdgcn_mfma_f32_32x32x2f32(
-              afrag[s], s_rt[2 * s + half][ct * 32 + l31], acc, 0, 0, 0);
-        const float rncol = s_rn[ct * 32 + l31];
-        // spill this subtile's keys to the wave's key matrix [col][row]
+          for (int s = 0; s < 6; ++s)
+            acc = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                afrag[s], s_rt[2 * s + half][ct * 32 + l31], acc, 0, 0, 0);
+          const float rncol = s_rn[ct * 32 + l31];
 #pragma unroll
-        for (int g = 0; g < 16; ++g) {
-          float key = fmaf(-2.f, acc[g], rncol);
-          s_km[wave][l31 * KM_PITCH + km_rowmap(g, half)] = key;
-        }
-        __builtin_amdgcn_wave_barrier();  // wave-internal: writes before reads
-        // owner scan: lane pair (l31, l31+32) streams row l31's 16 keys
-        const int row = l31;
-        const long long colbase = tb + ct * 32 + half * 16;
-#pragma unroll
-        for (int cc = 0; cc < 16; ++cc) {
-          float key = s_km[wave][(half * 16 + cc) * KM_PITCH + row];
-          if (key < wkey[qti]) {
-            // replace current worst, recompute worst (rare)
-            int ws = 0;
-            float wv = -FLT_MAX;
-#pragma unroll
-            for (int j = 0; j < KM_KMAX; ++j)
-              if (j < k && lk[qti][j] > wv) {
-                wv = lk[qti][j];
-                ws = j;
-              }
-            lk[qti][ws] = key;
-            li[qti][ws] = (int)(colbase + cc);
-            wv = -FLT_MAX;
-#pragma unroll
-            for (int j = 0; j < KM_KMAX; ++j)
-              if (j < k && lk[qti][j] > wv) wv = lk[qti][j];
-            wkey[qti] = wv;
+          for (int g = 0; g < 16; ++g) {
+            float key = fmaf(-2.f, acc[g], rncol);
+            s_km[wave][ct & 1][l31 * KM_PITCH + km_rowmap(g, half)] = key;
           }
         }
-        __builtin_amdgcn_wave_barrier();  // reads before next subtile's writes
+        __builtin_amdgcn_wave_barrier();
+        if (ct > 0) {
+          const int p = ct - 1;
+          // owner scan: lane pair (l31, l31+32) streams row l31's 16 keys
+          const int row = l31;
+          const long long colbase = tb + p * 32 + half * 16;
+#pragma unroll
+          for (int cc = 0; cc < 16; ++cc) {
+            float key = s_km[wave][p & 1][(half * 16 + cc) * KM_PITCH + row];
+            if (key < wkey[qti]) {
+              // replace current worst, recompute worst (rare)
+              int ws = 0;
+              float wv = -FLT_MAX;
+#pragma unroll
+              for (int j = 0; j < KM_KMAX; ++j)
+                if (j < k && lk[qti][j] > wv) {
+                  wv = lk[qti][j];
+                  ws = j;
+                }
+              lk[qti][ws] = key;
+              li[qti][ws] = (int)(colbase + cc);
+              wv = -FLT_MAX;
+#pragma unroll
+              for (int j = 0; j < KM_KMAX; ++j)
+                if (j < k && lk[qti][j] > wv) wv = lk[qti][j];
+              wkey[qti] = wv;
+            }
+          }
+        }
+        __builtin_amdgcn_wave_barrier();
       }
     }
   }
@@ -198,7 +207,7 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
 
   // ---- merge lane pairs, refine with exact distances, emit sorted --------
 #pragma unroll
-  for (int qti = 0; qti < 3; ++qti) {
+  for (int qti = 0; qti < 2; ++qti) {
     const int qt = wave + 4 * qti;
     const long long q = qb0 + qt * 32 + l31;
     // pull the partner half's sub-list (lane r+32 -> lane r and vice versa;

@@ -143,3 +143,44 @@ def test_cli_rejects_unknown_subcommand():
         cwd=REPO,
     )
     assert r.returncode != 0
+
+
+def test_stats_flag_emits_json_metrics():
+    import io
+    import json
+
+    from traffic_classifier_sdn_amd.flow.replay import TelemetryReplaySource
+    from traffic_classifier_sdn_amd.models import GaussianNB
+    from traffic_classifier_sdn_amd.serve import RealtimeClassifier
+    from traffic_classifier_sdn_amd.utils.datasets import load_reference_dataset
+
+    X, y = load_reference_dataset()
+    model = GaussianNB(device="cpu").fit(X[:500].astype("float64"), y[:500])
+    out, err = io.StringIO(), io.StringIO()
+    rc = RealtimeClassifier(model, out=out, stats=True, stats_out=err)
+    rc.run(TelemetryReplaySource(seed=1).stream(20))
+    lines = [json.loads(l) for l in err.getvalue().splitlines()]
+    assert lines, "no stats lines emitted"
+    assert lines[-1]["flows"] > 0 and lines[-1]["predict_ms"] > 0
+
+
+def test_kmeans_serve_uses_learned_cluster_names():
+    import io
+
+    import numpy as np
+
+    from traffic_classifier_sdn_amd.flow.replay import TelemetryReplaySource
+    from traffic_classifier_sdn_amd.models import KMeans
+    from traffic_classifier_sdn_amd.serve import RealtimeClassifier
+    from traffic_classifier_sdn_amd.utils.datasets import load_reference_dataset
+
+    X, _ = load_reference_dataset()
+    km = KMeans(n_clusters=6, device="cpu").fit(X[:2000])
+    km.cluster_label_names_ = np.asarray(
+        ["c0", "c1", "c2", "c3", "c4", "c5"], dtype=object
+    )
+    out = io.StringIO()
+    rc = RealtimeClassifier(km, out=out)
+    rc.run(TelemetryReplaySource(seed=1).stream(20))
+    body = out.getvalue()
+    assert any(f"c{i}" in body for i in range(6))

@@ -13,7 +13,9 @@ captured (ops.gpu) to amortise launch overhead at high poll rates.
 
 from __future__ import annotations
 
+import json
 import sys
+import time
 from typing import Iterable, Optional, TextIO
 
 import numpy as np
@@ -55,11 +57,15 @@ class RealtimeClassifier:
         model: Estimator,
         predict_every: int = 10,
         out: TextIO = sys.stdout,
+        stats: bool = False,
+        stats_out: TextIO = sys.stderr,
     ) -> None:
         self.model = model
         self.parser = PollStreamParser()
         self.predict_every = predict_every
         self.out = out
+        self.stats = stats
+        self.stats_out = stats_out
         self._last_batch = 0
 
     def classify_now(self) -> np.ndarray:
@@ -69,7 +75,14 @@ class RealtimeClassifier:
         X = table.feature_matrix(dtype=np.float32)
         pred = self.model.predict(X)
         if self.model.classes_ is None:  # unsupervised: cluster ids -> names
-            pred = map_cluster_labels(pred)
+            names = getattr(self.model, "cluster_label_names_", None)
+            if names is not None:
+                # mode-based map learned at fit time (fit.py); the reference's
+                # fixed index->name map silently mislabels
+                idx = np.asarray(pred).ravel().astype(np.int64)
+                pred = np.asarray(names, dtype=object)[idx]
+            else:
+                pred = map_cluster_labels(pred)
         return pred
 
     def feed(self, line) -> bool:
@@ -85,9 +98,26 @@ class RealtimeClassifier:
             return False
         if (self.parser.records - self._last_batch) >= self.predict_every:
             self._last_batch = self.parser.records
+            t0 = time.perf_counter()
             labels = self.classify_now()
+            predict_s = time.perf_counter() - t0
             self.out.write(render_flow_table(self.parser.table, labels) + "\n")
             self.out.flush()
+            if self.stats:
+                n = len(self.parser.table)
+                self.stats_out.write(
+                    json.dumps(
+                        {
+                            "flows": n,
+                            "records": self.parser.records,
+                            "predict_ms": predict_s * 1e3,
+                            "flows_per_sec": n / predict_s if predict_s > 0 else None,
+                            "device": str(getattr(self.model, "device", "cpu")),
+                        }
+                    )
+                    + "\n"
+                )
+                self.stats_out.flush()
             return True
         return False
 

@@ -97,19 +97,20 @@ __global__ void logistic_grad_kernel(const double* __restrict__ X,
       logits[c] = s;
       m = fmax(m, s);
     }
+    int yc = (int)y[row];
     double z = 0.0;
+    // keep logit_y - m BEFORE the exp: log(exp(t)) underflows to -inf for
+    // t < -745, which raw-scale flow features reach easily once the weights
+    // grow — the CPU oracle's log_softmax never exponentiates the margin
+    double ly = 0.0;
 #pragma unroll
     for (int c = 0; c < C; ++c) {
-      logits[c] = exp(logits[c] - m);
+      double t = logits[c] - m;
+      if (c == yc) ly = t;
+      logits[c] = exp(t);
       z += logits[c];
     }
-    int yc = (int)y[row];
-    // loss = -(logit_y - m - log z); logits[yc] read via unrolled select
-    double py = 0.0;
-#pragma unroll
-    for (int c = 0; c < C; ++c)
-      if (c == yc) py = logits[c];
-    local_loss += -(log(py) - log(z));
+    local_loss += -(ly - log(z));
     double inv_z = 1.0 / z;
 #pragma unroll
     for (int c = 0; c < C; ++c) {

@@ -68,10 +68,14 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
   __shared__ float s_qt[KM_F][KM_QB];            // centered Q^T
   __shared__ float s_rt[KM_F][KM_TB];            // centered R-tile^T
   __shared__ float s_rn[KM_TB];                  // ||r-c||^2 (FLT_MAX pad)
-  // per-wave DOUBLE-BUFFERED key matrix [col][row]: the owner scan of
-  // subtile ct-1 overlaps the MFMA + key spill of subtile ct, so the LDS
-  // write->read dependency never stalls the wave
-  __shared__ float s_km[4][2][32 * KM_PITCH];
+  // per-wave survivor key matrix [col][row] + per-row survivor bitmask:
+  // producers only spill keys that beat the (stale-tolerant) per-row
+  // threshold, so in steady state a subtile costs 16 broadcast threshold
+  // reads + compares and the owner scan reads nothing at all — the LDS
+  // pipe stops being the bottleneck (was: full 16-write/16-read key spill)
+  __shared__ float s_km[4][32 * KM_PITCH];
+  __shared__ unsigned s_mask[4][32];
+  __shared__ float s_worst[KM_QB];  // per-row k-th-best key (owners update)
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -104,7 +108,11 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
     Row12 x = load_row12(Q, q);
 #pragma unroll
     for (int j = 0; j < KM_F; ++j) s_qt[j][i] = x.v[j] - cm[j];
+    s_worst[i] = FLT_MAX;
   }
+  if (tid < 4)
+#pragma unroll
+    for (int r = 0; r < 32; ++r) s_mask[tid][r] = 0u;
   __syncthreads();
 
   // per-(lane, query-tile) top-k sub-list in registers; the lane pair
@@ -153,34 +161,44 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
       float afrag[6];
 #pragma unroll
       for (int s = 0; s < 6; ++s) afrag[s] = s_qt[2 * s + half][rowbase + l31];
+      // per-row threshold cache (broadcast reads); stale values only admit
+      // extra survivors — the owner re-checks against its exact register
+      // worst — never miss one
+      float tau[16];
+#pragma unroll
+      for (int g = 0; g < 16; ++g) tau[g] = s_worst[rowbase + km_rowmap(g, half)];
 
-      // software pipeline: iteration ct spills subtile ct's keys into
-      // buffer ct&1 and scans subtile ct-1 from the other buffer
-      for (int ct = 0; ct <= KM_TB / 32; ++ct) {
-        if (ct < KM_TB / 32) {
-          f32x16 acc = {};
+      for (int ct = 0; ct < KM_TB / 32; ++ct) {
+        f32x16 acc = {};
 #pragma unroll
-          for (int s = 0; s < 6; ++s)
-            acc = __builtin_amdgcn_mfma_f32_32x32x2f32(
-                afrag[s], s_rt[2 * s + half][ct * 32 + l31], acc, 0, 0, 0);
-          const float rncol = s_rn[ct * 32 + l31];
+        for (int s = 0; s < 6; ++s)
+          acc = __builtin_amdgcn_mfma_f32_32x32x2f32(
+              afrag[s], s_rt[2 * s + half][ct * 32 + l31], acc, 0, 0, 0);
+        const float rncol = s_rn[ct * 32 + l31];
+        // survivors spill (key -> own slot, col bit -> row mask); padded
+        // columns carry key = FLT_MAX and never pass
 #pragma unroll
-          for (int g = 0; g < 16; ++g) {
-            float key = fmaf(-2.f, acc[g], rncol);
-            s_km[wave][ct & 1][l31 * KM_PITCH + km_rowmap(g, half)] = key;
+        for (int g = 0; g < 16; ++g) {
+          float key = fmaf(-2.f, acc[g], rncol);
+          if (key < tau[g]) {
+            int row = km_rowmap(g, half);
+            s_km[wave][l31 * KM_PITCH + row] = key;
+            atomicOr(&s_mask[wave][row], 1u << l31);
           }
         }
         __builtin_amdgcn_wave_barrier();
-        if (ct > 0) {
-          const int p = ct - 1;
-          // owner scan: lane pair (l31, l31+32) streams row l31's 16 keys
+        // owner scan: lane pair (l31, l31+32) splits the survivor bits of
+        // row l31 (low/high 16 columns)
+        {
           const int row = l31;
-          const long long colbase = tb + p * 32 + half * 16;
-#pragma unroll
-          for (int cc = 0; cc < 16; ++cc) {
-            float key = s_km[wave][p & 1][(half * 16 + cc) * KM_PITCH + row];
+          unsigned m = s_mask[wave][row];
+          unsigned mh = half ? (m >> 16) : (m & 0xffffu);
+          const long long colbase = tb + ct * 32 + half * 16;
+          while (mh) {
+            int cc = __ffs(mh) - 1;
+            mh &= mh - 1;
+            float key = s_km[wave][(half * 16 + cc) * KM_PITCH + row];
             if (key < wkey[qti]) {
-              // replace current worst, recompute worst (rare)
               int ws = 0;
               float wv = -FLT_MAX;
 #pragma unroll
@@ -197,6 +215,17 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
                 if (j < k && lk[qti][j] > wv) wv = lk[qti][j];
               wkey[qti] = wv;
             }
+          }
+          // a produced column can land in either half's sub-list, so the
+          // shared threshold must be conservative for BOTH: publish the max
+          // of the lane pair's worsts (stale/loose admits extra survivors,
+          // never drops one)
+          float other = __shfl(wkey[qti], lane ^ 32, WAVE);
+          float pub = fmaxf(wkey[qti], other);
+          __builtin_amdgcn_wave_barrier();
+          if (half == 0) {
+            s_worst[rowbase + row] = pub;
+            if (m) s_mask[wave][row] = 0u;
           }
         }
         __builtin_amdgcn_wave_barrier();

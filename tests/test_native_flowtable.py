@@ -1,5 +1,7 @@
 """C++ flow table parity with the Python FlowTable on identical streams."""
 
+import os
+
 import numpy as np
 import pytest
 
@@ -58,3 +60,23 @@ def test_bulk_vs_per_line():
     b = native.NativePollParser()
     b.feed_buffer("\n".join(lines))
     np.testing.assert_allclose(b.table.feature_matrix(), a.table.feature_matrix())
+
+
+def test_sanitizer_harness(tmp_path):
+    """Build the pure-C++ core under ASan+UBSan and run the adversarial /
+    randomized / semantic property harness (SURVEY.md §5 sanitizer parity)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    src = os.path.join(repo, "tools", "flowtable_san.cpp")
+    exe = str(tmp_path / "flowtable_san")
+    build = subprocess.run(
+        ["g++", "-std=c++17", "-O1", "-g", "-fsanitize=address,undefined",
+         "-fno-sanitize-recover=all", src, "-o", exe],
+        capture_output=True, text=True, timeout=180,
+    )
+    assert build.returncode == 0, build.stderr
+    run = subprocess.run([exe], capture_output=True, text=True, timeout=120)
+    assert run.returncode == 0, run.stdout + run.stderr
+    assert "SANITIZER HARNESS OK" in run.stdout

@@ -44,3 +44,9 @@ DEV double wave_sum(double x) {
   for (int off = 32; off > 0; off >>= 1) x += __shfl_down(x, off, 64);
   return x;
 }
+
+DEV unsigned long long wave_sum(unsigned long long x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_down(x, off, 64);
+  return x;
+}

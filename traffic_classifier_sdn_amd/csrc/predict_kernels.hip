@@ -282,11 +282,79 @@ __global__ void rf_predict_kernel(const float* __restrict__ X,
   }
 }
 
+// Small-batch variant: ONE WAVE PER ROW, trees split across lanes.  At
+// serve batch sizes the row-per-lane kernel runs a ~900-step serial
+// dependent-load chain per row on a mostly idle chip; here each lane walks
+// only ceil(T/64) trees and n waves fill the SIMDs.  Nodes are read from
+// global memory — the packed forest (tens of KB) stays L2-resident.
+template <int C>
+__launch_bounds__(256) __global__ void rf_predict_wave_kernel(
+    const float* __restrict__ X, const uint2* __restrict__ nodes,
+    const int* __restrict__ roots, const float* __restrict__ leaf_proba,
+    int* __restrict__ out, long long n, int T) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long long row = (long long)blockIdx.x * (blockDim.x / WAVE) +
+                        (threadIdx.x >> 6);
+  if (row >= n) return;
+  Row12 x = load_row12(X, row);  // broadcast load
+  float acc[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) acc[c] = 0.f;
+  unsigned long long votes = 0ull;  // 10-bit packed per-class pure counts
+  for (int t = lane; t < T; t += WAVE) {
+    int idx = roots[t];
+    while (true) {
+      uint2 node = nodes[idx];
+      unsigned feat = node.y & 0xffu;
+      if (feat >= 0xf0u) {
+        if (feat == 0xffu) {
+          int pr = (int)node.x * C;
+#pragma unroll
+          for (int c = 0; c < C; ++c) acc[c] += leaf_proba[pr + c];
+        } else {
+          votes += 1ull << ((feat & 0xfu) * 10);
+        }
+        break;
+      }
+      float thr = __uint_as_float(node.x);
+      idx = (sel12(x, feat) <= thr) ? idx + 1 : (int)(node.y >> 8);
+    }
+  }
+  votes = wave_sum(votes);  // per-field sums stay < 1024: no cross-field carry
+#pragma unroll
+  for (int c = 0; c < C; ++c) acc[c] = wave_sum(acc[c]);
+  if (lane == 0) {
+    float best = -INFINITY;
+    int bi = 0;
+#pragma unroll
+    for (int c = 0; c < C; ++c) {
+      float sc = acc[c] + (float)((votes >> (c * 10)) & 1023ull);
+      if (sc > best) { best = sc; bi = c; }
+    }
+    out[row] = bi;
+  }
+}
+
 extern "C" void launch_rf_predict(const float* X, const unsigned* nodes,
                                   const int* roots, const float* leaf_proba,
                                   int* out, long long n, int n_nodes,
                                   int n_leaves, int T, int C,
                                   hipStream_t stream) {
+  if (n <= 32768) {  // wave-per-row fills the chip at serve batch sizes
+    dim3 wgrid((unsigned)((n + 3) / 4));
+#define RFW_CASE(CV)                                                        \
+  case CV:                                                                  \
+    hipLaunchKernelGGL((rf_predict_wave_kernel<CV>), wgrid, dim3(256), 0,   \
+                       stream, X, reinterpret_cast<const uint2*>(nodes),    \
+                       roots, leaf_proba, out, n, T);                       \
+    return;
+    switch (C) {
+      RFW_CASE(2) RFW_CASE(3) RFW_CASE(4) RFW_CASE(5) RFW_CASE(6) RFW_CASE(7)
+      RFW_CASE(8)
+      default: break;
+    }
+#undef RFW_CASE
+  }
   const int block = 512;
   // Occupancy first: with nodes-only LDS a 43 KB forest admits 3 blocks
   // (24 waves) per CU.  Leaf probabilities go to LDS only when everything
@@ -413,6 +481,74 @@ __global__ void svc_predict_kernel(const float* __restrict__ X,
   }
 }
 
+// Small-batch variant: ONE WAVE PER ROW, SVs split across lanes.  At serve
+// batch sizes (thousands of rows) the row-per-lane kernel leaves most of
+// the chip idle and each lane walks all nsv SVs serially; here n waves fill
+// the 1024 SIMDs and the per-row latency drops ~100x (serve-path profile:
+// profiles/serve_kernel_trace_r01.md).  SV rows stream from global memory —
+// the working set (nsv * 69 B) stays L2-resident and is shared by every
+// wave.  The per-SV class is NOT wave-uniform here, so the OVO accumulator
+// update is an unrolled compile-time switch on the class id.
+template <int C>
+__launch_bounds__(256) __global__ void svc_predict_wave_kernel(
+    const float* __restrict__ X, const float* __restrict__ SV,
+    const float* __restrict__ dual, const unsigned char* __restrict__ svclass,
+    const float* __restrict__ intercept, int* __restrict__ out, long long n,
+    int nsv, float gamma) {
+  constexpr int F = 12;
+  constexpr int CR = C - 1;
+  constexpr int NPAIR = C * (C - 1) / 2;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long long row = (long long)blockIdx.x * (blockDim.x / WAVE) +
+                        (threadIdx.x >> 6);
+  if (row >= n) return;
+  Row12 x = load_row12(X, row);  // broadcast load, all lanes same row
+  float acc[C * CR];
+#pragma unroll
+  for (int i = 0; i < C * CR; ++i) acc[i] = 0.f;
+  for (int s = lane; s < nsv; s += WAVE) {
+    Row12 sv = load_row12(SV, s);
+    float d = 0.f;
+#pragma unroll
+    for (int j = 0; j < F; ++j) {
+      float t = x.v[j] - sv.v[j];
+      d = fmaf(t, t, d);
+    }
+    float kv = __expf(-gamma * d);
+    int c = svclass[s];
+#pragma unroll
+    for (int cc = 0; cc < C; ++cc) {
+      if (c == cc) {
+#pragma unroll
+        for (int r = 0; r < CR; ++r)
+          acc[cc * CR + r] = fmaf(dual[(long long)r * nsv + s], kv, acc[cc * CR + r]);
+      }
+    }
+  }
+  // cross-lane reduction of the C*(C-1) partial sums
+#pragma unroll
+  for (int i = 0; i < C * CR; ++i) acc[i] = wave_sum(acc[i]);
+  if (lane == 0) {
+    int votes[C];
+#pragma unroll
+    for (int c = 0; c < C; ++c) votes[c] = 0;
+    int p = 0;
+#pragma unroll
+    for (int i = 0; i < C; ++i)
+#pragma unroll
+      for (int j = i + 1; j < C; ++j, ++p) {
+        float dec = acc[i * CR + (j - 1)] + acc[j * CR + i] + intercept[p];
+        if (dec > 0.f) votes[i]++; else votes[j]++;
+      }
+    int best = -1, bi = 0;
+#pragma unroll
+    for (int c = 0; c < C; ++c)
+      if (votes[c] > best) { best = votes[c]; bi = c; }
+    out[row] = bi;
+    (void)NPAIR;
+  }
+}
+
 extern "C" void launch_svc_predict(const float* X, const float* SV,
                                    const float* dual,
                                    const unsigned char* svclass,
@@ -420,6 +556,20 @@ extern "C" void launch_svc_predict(const float* X, const float* SV,
                                    long long n, int nsv, int C, float gamma,
                                    hipStream_t stream) {
   const int block = 256;
+  if (n <= 32768) {  // wave-per-row fills the chip at serve batch sizes
+    dim3 wgrid((unsigned)((n + 3) / 4));
+#define SVC_WCASE(CV)                                                       \
+  case CV:                                                                  \
+    hipLaunchKernelGGL((svc_predict_wave_kernel<CV>), wgrid, dim3(block),   \
+                       0, stream, X, SV, dual, svclass, intercept, out, n,  \
+                       nsv, gamma);                                         \
+    return;
+    switch (C) {
+      SVC_WCASE(2) SVC_WCASE(3) SVC_WCASE(4) SVC_WCASE(5) SVC_WCASE(6)
+      default: break;
+    }
+#undef SVC_WCASE
+  }
   dim3 grid(ts_grid(n, block));
 #define SVC_CASE(CV)                                                        \
   case CV:                                                                  \

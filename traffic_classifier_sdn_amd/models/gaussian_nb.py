@@ -68,9 +68,9 @@ class GaussianNB(Estimator):
         Xt = as_tensor(X, self.device, torch.float32)
         return ops.gnb_argmax(
             Xt,
-            self.theta_.to(Xt.dtype),
-            self.var_.to(Xt.dtype),
-            self.class_prior_.to(Xt.dtype),
+            self.theta_,
+            self.var_,
+            self.class_prior_,
         )
 
     def joint_log_likelihood(self, X: ArrayLike) -> torch.Tensor:

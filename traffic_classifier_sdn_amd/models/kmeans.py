@@ -120,7 +120,7 @@ class KMeans(Estimator):
 
     def predict_index(self, X: ArrayLike) -> torch.Tensor:
         Xt = as_tensor(X, self.device, torch.float32)
-        return ops.kmeans_labels(Xt, self.cluster_centers_.to(Xt.dtype))
+        return ops.kmeans_labels(Xt, self.cluster_centers_)
 
     # -- checkpointing -------------------------------------------------
     def to_params(self) -> Dict[str, Any]:

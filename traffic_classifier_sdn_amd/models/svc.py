@@ -143,11 +143,13 @@ class SVC(Estimator):
                 self.n_support_.to(Xt.device),
             ).to(torch.uint8).contiguous()
             self._svclass = sc
+        # raw (f64) params: the op layer caches the f32 copies so the
+        # hipGraph serve path replays no conversion kernels
         return ops.svc_predict(
             Xt,
-            self.support_vectors_.to(Xt.dtype),
-            self.dual_coef_.to(Xt.dtype),
-            self.intercept_.to(Xt.dtype),
+            self.support_vectors_,
+            self.dual_coef_,
+            self.intercept_,
             self.n_support_,
             self.gamma_,
             svclass=sc,

@@ -23,7 +23,7 @@ import torch
 def linear_logits(X: torch.Tensor, coef: torch.Tensor, intercept: torch.Tensor) -> torch.Tensor:
     """logits[n,c] = X @ coef^T + intercept  (reference sklearn
     LogisticRegression.decision_function)."""
-    return X @ coef.T + intercept
+    return X @ coef.T.to(X.dtype) + intercept.to(X.dtype)
 
 
 def linear_argmax(X: torch.Tensor, coef: torch.Tensor, intercept: torch.Tensor) -> torch.Tensor:
@@ -64,6 +64,9 @@ def gnb_joint_loglik(
     X: torch.Tensor, theta: torch.Tensor, var: torch.Tensor, class_prior: torch.Tensor
 ) -> torch.Tensor:
     """Per-class joint log-likelihood (sklearn GaussianNB._joint_log_likelihood)."""
+    theta = theta.to(X.dtype)
+    var = var.to(X.dtype)
+    class_prior = class_prior.to(X.dtype)
     # const[c] = log prior[c] - 0.5 * sum_j log(2*pi*var[c,j])
     const = torch.log(class_prior) - 0.5 * torch.log(2.0 * torch.pi * var).sum(dim=1)
     # quad[n,c] = -0.5 * sum_j (x[n,j]-theta[c,j])^2 / var[c,j]
@@ -110,6 +113,7 @@ def kmeans_assign(
     X: torch.Tensor, centers: torch.Tensor
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
     """Lloyd assignment step. Returns (labels, counts[K], sums[K,F], inertia)."""
+    centers = centers.to(X.dtype)
     d = pairwise_sqdist(X, centers)
     dmin, labels = torch.min(d, dim=1)
     K, F = centers.shape
@@ -218,8 +222,8 @@ def svc_predict(
     gamma: float,
     svclass: torch.Tensor = None,
 ) -> torch.Tensor:
-    K = rbf_kernel(X, SV, gamma)
-    dec = svc_ovo_decision(K, dual_coef, intercept, n_support)
+    K = rbf_kernel(X, SV.to(X.dtype), gamma)
+    dec = svc_ovo_decision(K, dual_coef.to(X.dtype), intercept.to(X.dtype), n_support)
     return svc_vote(dec, int(n_support.numel()))
 
 

@@ -48,6 +48,11 @@ void launch_flow_features(const double*, const double*, const double*, float*,
                           long long, hipStream_t);
 void launch_smo_select(const float*, const double*, const double*, double,
                        long long, unsigned long long*, hipStream_t);
+void launch_smo_solve(const float*, const float*, double*, const double*,
+                      unsigned long long*, float*, double*, double, double,
+                      float, hipStream_t);
+void launch_smo_update_dev(const float*, const float*, double*, const float*,
+                           const double*, float, long long, hipStream_t);
 void launch_smo_update(const float*, const float*, double*, const float*,
                        double, double, float, long long, hipStream_t);
 }
@@ -273,6 +278,36 @@ static void smo_select(torch::Tensor y, torch::Tensor alpha,
                     cur_stream());
 }
 
+static void smo_solve(torch::Tensor X, torch::Tensor y, torch::Tensor alpha,
+                      torch::Tensor grad, torch::Tensor sel, torch::Tensor rows,
+                      torch::Tensor sol, double C, double tol, double gamma) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(y, torch::kFloat32);
+  CHECK_IN(alpha, torch::kFloat64);
+  CHECK_IN(grad, torch::kFloat64);
+  CHECK_IN(sel, torch::kInt64);
+  CHECK_IN(rows, torch::kFloat32);
+  CHECK_IN(sol, torch::kFloat64);
+  launch_smo_solve(X.data_ptr<float>(), y.data_ptr<float>(),
+                   alpha.data_ptr<double>(), grad.data_ptr<double>(),
+                   reinterpret_cast<unsigned long long*>(sel.data_ptr<int64_t>()),
+                   rows.data_ptr<float>(), sol.data_ptr<double>(), C, tol,
+                   (float)gamma, cur_stream());
+}
+
+static void smo_update_dev(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
+                           torch::Tensor rows, torch::Tensor sol, double gamma) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(y, torch::kFloat32);
+  CHECK_IN(grad, torch::kFloat64);
+  CHECK_IN(rows, torch::kFloat32);
+  CHECK_IN(sol, torch::kFloat64);
+  launch_smo_update_dev(X.data_ptr<float>(), y.data_ptr<float>(),
+                        grad.data_ptr<double>(), rows.data_ptr<float>(),
+                        sol.data_ptr<double>(), (float)gamma, X.size(0),
+                        cur_stream());
+}
+
 static void smo_update(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
                        torch::Tensor rows, double yidai, double yjdaj,
                        double gamma) {
@@ -288,6 +323,8 @@ static void smo_update(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("smo_select", &smo_select, "WSS-1 pair candidate selection");
   m.def("smo_update", &smo_update, "fused RBF-row gradient update");
+  m.def("smo_solve", &smo_solve, "device-side SMO pair solve (fused iteration)");
+  m.def("smo_update_dev", &smo_update_dev, "gradient update from device sol buffer");
   m.def("gnb_predict", &gnb_predict, "fused GaussianNB loglik+argmax");
   m.def("linear_argmax", &linear_argmax, "logits+argmax");
   m.def("kmeans_assign", &kmeans_assign, "Lloyd assign + partial update");

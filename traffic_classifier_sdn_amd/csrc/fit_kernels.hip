@@ -253,6 +253,85 @@ extern "C" void launch_smo_select(const float* y, const double* alpha,
                      stream, y, alpha, grad, C, n, out);
 }
 
+// Device-side analytic 2-variable solve (single-GPU fast path): decodes the
+// select result, solves the pair subproblem, updates alpha IN PLACE, stages
+// the two rows + deltas for the gradient kernel, and re-zeroes the select
+// buffer — so a whole SMO iteration runs select→solve→update with NO host
+// round-trip; the host polls the status word once per iteration chunk.
+// sol layout (f64[4]): [yidai, yjdaj, status(0 run/1 converged), gap]
+__global__ void smo_solve_kernel(const float* __restrict__ X,
+                                 const float* __restrict__ y,
+                                 double* __restrict__ alpha,
+                                 const double* __restrict__ grad,
+                                 unsigned long long* __restrict__ sel,
+                                 float* __restrict__ rows,  // [24]
+                                 double* __restrict__ sol,  // [4]
+                                 double C, double tol, float gamma) {
+  if (threadIdx.x != 0) return;  // tiny scalar epilogue: one lane suffices
+  if (sol[2] != 0.0) return;     // already converged: stay converged
+  unsigned long long pu = sel[0], pl = sel[1];
+  sel[0] = 0;
+  sel[1] = 0;  // re-arm the atomicMax select for the next iteration
+  if (pu == 0 || pl == 0) {
+    sol[0] = sol[1] = 0.0;
+    sol[2] = 1.0;
+    return;
+  }
+  int i = (int)(pu & 0xffffffffull);
+  int j = (int)(pl & 0xffffffffull);
+  double yi = (double)y[i], yj = (double)y[j];
+  double ai = alpha[i], aj = alpha[j];
+  double gi = grad[i], gj = grad[j];
+  double up_val = -yi * gi;
+  double low_val = -yj * gj;
+  double gap = up_val - low_val;
+  sol[3] = gap;
+  if (gap < tol) {
+    sol[0] = sol[1] = 0.0;
+    sol[2] = 1.0;
+    return;
+  }
+  double d2 = 0.0;
+#pragma unroll
+  for (int k = 0; k < 12; ++k) {
+    float xi = X[(long long)i * 12 + k];
+    float xj = X[(long long)j * 12 + k];
+    rows[k] = xi;
+    rows[12 + k] = xj;
+    double t = (double)xi - (double)xj;
+    d2 += t * t;
+  }
+  double kij = exp(-(double)gamma * d2);
+  double a = 2.0 - 2.0 * yi * yj * kij;
+  if (a <= 0.0) a = 1e-12;
+  double d = gap / a;
+  double ai_new = ai + yi * d;
+  double s = yi * ai + yj * aj;
+  ai_new = fmin(fmax(ai_new, 0.0), C);
+  double aj_new = yj * (s - yi * ai_new);
+  aj_new = fmin(fmax(aj_new, 0.0), C);
+  ai_new = yi * (s - yj * aj_new);
+  ai_new = fmin(fmax(ai_new, 0.0), C);
+  double dai = ai_new - ai, daj = aj_new - aj;
+  if (fabs(dai) < 1e-16 && fabs(daj) < 1e-16) {
+    sol[0] = sol[1] = 0.0;
+    sol[2] = 1.0;
+    return;
+  }
+  alpha[i] = ai_new;
+  alpha[j] = aj_new;
+  sol[0] = yi * dai;
+  sol[1] = yj * daj;
+}
+
+extern "C" void launch_smo_solve(const float* X, const float* y, double* alpha,
+                                 const double* grad, unsigned long long* sel,
+                                 float* rows, double* sol, double C, double tol,
+                                 float gamma, hipStream_t stream) {
+  hipLaunchKernelGGL(smo_solve_kernel, dim3(1), dim3(64), 0, stream, X, y,
+                     alpha, grad, sel, rows, sol, C, tol, gamma);
+}
+
 // grad[t] += y_t * (yi*dai*K(xi,x_t) + yj*daj*K(xj,x_t)); the two RBF rows
 // are computed on the fly (fused — no kernel matrix is ever materialised).
 // xi/xj come from the 24-float staging buffer `rows` (filled by the host
@@ -282,6 +361,48 @@ __global__ void smo_update_kernel(const float* __restrict__ X,
     double kj = (double)__expf(-gamma * dj);
     grad[t] += (double)y[t] * (yidai * ki + yjdaj * kj);
   }
+}
+
+// Variant driven by the device-resident sol buffer (smo_solve output):
+// deltas come from sol[0..1]; a converged status (sol[2] != 0) no-ops.
+__global__ void smo_update_dev_kernel(const float* __restrict__ X,
+                                      const float* __restrict__ y,
+                                      double* __restrict__ grad,
+                                      const float* __restrict__ rows,  // [24]
+                                      const double* __restrict__ sol,  // [4]
+                                      float gamma, long long n) {
+  __shared__ float s_rows[24];
+  __shared__ double s_d[2];
+  if (threadIdx.x < 24) s_rows[threadIdx.x] = rows[threadIdx.x];
+  if (threadIdx.x < 2) s_d[threadIdx.x] = sol[threadIdx.x];
+  __syncthreads();
+  if (sol[2] != 0.0) return;
+  const double yidai = s_d[0], yjdaj = s_d[1];
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < n;
+       t += stride) {
+    Row12 x = load_row12(X, t);
+    float di = 0.f, dj = 0.f;
+#pragma unroll
+    for (int k = 0; k < 12; ++k) {
+      float a = x.v[k] - s_rows[k];
+      float b = x.v[k] - s_rows[12 + k];
+      di = fmaf(a, a, di);
+      dj = fmaf(b, b, dj);
+    }
+    double ki = (double)__expf(-gamma * di);
+    double kj = (double)__expf(-gamma * dj);
+    grad[t] += (double)y[t] * (yidai * ki + yjdaj * kj);
+  }
+}
+
+extern "C" void launch_smo_update_dev(const float* X, const float* y,
+                                      double* grad, const float* rows,
+                                      const double* sol, float gamma,
+                                      long long n, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(smo_update_dev_kernel, dim3(ts_grid(n, block)),
+                     dim3(block), 0, stream, X, y, grad, rows, sol, gamma, n);
 }
 
 extern "C" void launch_smo_update(const float* X, const float* y, double* grad,

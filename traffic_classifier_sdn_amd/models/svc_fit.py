@@ -62,6 +62,66 @@ def _grad_update_cpu(X, y, grad, xi, xj, yidai, yjdaj, gamma):
     grad += y.double() * (yidai * ki + yjdaj * kj)
 
 
+def _smo_fused_gpu(X, y, alpha, grad, C, gamma, tol, max_iter, chunk=128):
+    """Single-GPU fast path: the whole select→solve→update iteration runs on
+    device (csrc smo_solve updates alpha in place and re-arms the select
+    buffer), so the host only polls the convergence status once per
+    ``chunk`` iterations instead of 5 round-trips per iteration."""
+    from ..ops import gpu as og
+
+    device = X.device
+    sel = torch.zeros(2, dtype=torch.int64, device=device)
+    rows = torch.zeros(24, dtype=torch.float32, device=device)
+    sol = torch.zeros(4, dtype=torch.float64, device=device)
+
+    def one_iter():
+        og._ext.smo_select(y, alpha, grad, float(C), sel)
+        og._ext.smo_solve(X, y, alpha, grad, sel, rows, sol, float(C), float(tol), float(gamma))
+        og._ext.smo_update_dev(X, y, grad, rows, sol, float(gamma))
+
+    # capture `chunk` iterations in one hipGraph: every kernel argument is a
+    # device buffer, so the captured sequence is exact; a converged solve
+    # turns the remaining replayed iterations into no-ops (status latch)
+    graph = None
+    if max_iter < 8 * chunk:
+        # capture cost (~chunk x 3 launches) only amortises on long fits
+        it = 0
+        while it < max_iter:
+            for _ in range(min(chunk, max_iter - it)):
+                one_iter()
+                it += 1
+            if float(sol[2]) != 0.0:
+                break
+        return it
+    try:
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            one_iter()  # warmup outside capture
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            for _ in range(chunk):
+                one_iter()
+        it = 1  # the warmup iteration is real work
+    except Exception:
+        graph = None
+        it = 0
+
+    while it < max_iter:
+        if graph is not None and (max_iter - it) >= chunk:
+            graph.replay()
+            it += chunk
+        else:
+            for _ in range(min(chunk, max_iter - it)):
+                one_iter()
+                it += 1
+        if float(sol[2]) != 0.0:  # one D2H per chunk
+            break
+    return it
+
+
 def smo_fit_pair(
     X: torch.Tensor,
     y_pm: torch.Tensor,
@@ -86,6 +146,9 @@ def smo_fit_pair(
     rows_buf = torch.zeros(24, dtype=torch.float32, device=device) if is_gpu else None
     world = dist.world_size()
     it = 0
+    if is_gpu and world == 1:
+        it = _smo_fused_gpu(X.contiguous(), y.contiguous(), alpha, grad, C, gamma, tol, max_iter)
+        return alpha, _smo_intercept(y, alpha, grad, C, device), it
     for it in range(1, max_iter + 1):
         if is_gpu:
             i, up_val, j, low_val = _local_select_gpu(y, alpha, grad, C, sel_buf)
@@ -165,23 +228,25 @@ def smo_fit_pair(
         else:
             _grad_update_cpu(X, y, grad, xif, xjf, yi * dai, yj * daj, gamma)
 
-    # intercept from free vectors (global)
+    return alpha, _smo_intercept(y, alpha, grad, C, device), it
+
+
+def _smo_intercept(y, alpha, grad, C, device) -> float:
+    """Intercept from free vectors (global across ranks)."""
     myg = -(y.double() * grad)
     free = (alpha > 1e-12) & (alpha < C - 1e-12)
     ssum = torch.tensor([float(myg[free].sum()), float(free.sum())], dtype=torch.float64)
     dist.allreduce_(ssum)
     if float(ssum[1]) > 0:
-        b = float(ssum[0] / ssum[1])
-    else:
-        up = ((y > 0) & (alpha < C)) | ((y < 0) & (alpha > 0))
-        low = ((y > 0) & (alpha > 0)) | ((y < 0) & (alpha < C))
-        hi = torch.tensor(
-            [float(torch.where(up, myg, torch.tensor(-math.inf, dtype=torch.float64, device=device)).max())]
-        )
-        lo = torch.tensor(
-            [float(torch.where(low, myg, torch.tensor(math.inf, dtype=torch.float64, device=device)).min())]
-        )
-        dist.allreduce_(hi, op=torch.distributed.ReduceOp.MAX if dist.is_initialized() else None)
-        dist.allreduce_(lo, op=torch.distributed.ReduceOp.MIN if dist.is_initialized() else None)
-        b = float((hi[0] + lo[0]) / 2.0)
-    return alpha, b, it
+        return float(ssum[0] / ssum[1])
+    up = ((y > 0) & (alpha < C)) | ((y < 0) & (alpha > 0))
+    low = ((y > 0) & (alpha > 0)) | ((y < 0) & (alpha < C))
+    hi = torch.tensor(
+        [float(torch.where(up, myg, torch.tensor(-math.inf, dtype=torch.float64, device=device)).max())]
+    )
+    lo = torch.tensor(
+        [float(torch.where(low, myg, torch.tensor(math.inf, dtype=torch.float64, device=device)).min())]
+    )
+    dist.allreduce_(hi, op=torch.distributed.ReduceOp.MAX if dist.is_initialized() else None)
+    dist.allreduce_(lo, op=torch.distributed.ReduceOp.MIN if dist.is_initialized() else None)
+    return float((hi[0] + lo[0]) / 2.0)

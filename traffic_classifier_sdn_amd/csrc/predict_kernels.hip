@@ -251,24 +251,44 @@ __global__ void rf_predict_kernel(const float* __restrict__ X,
 #pragma unroll
     for (int c = 0; c < C; ++c) acc[c] = 0.f;
     unsigned long long votes = 0ull;  // 10-bit packed per-class pure counts
+    int decided = 0;
     for (int t = 0; t < T; ++t) {
-      int idx = roots[t];
-      while (true) {
-        uint2 node = LDS_NODES ? s_nodes[idx] : nodes[idx];
-        unsigned feat = node.y & 0xffu;
-        if (feat >= 0xf0u) {
-          if (feat == 0xffu) {  // mixed leaf: read the distribution
-            int pr = (int)node.x * C;
+      if (!decided) {
+        int idx = roots[t];
+        while (true) {
+          uint2 node = LDS_NODES ? s_nodes[idx] : nodes[idx];
+          unsigned feat = node.y & 0xffu;
+          if (feat >= 0xf0u) {
+            if (feat == 0xffu) {  // mixed leaf: read the distribution
+              int pr = (int)node.x * C;
 #pragma unroll
-            for (int c = 0; c < C; ++c)
-              acc[c] += LDS_PROBS ? s_probs[pr + c] : leaf_proba[pr + c];
-          } else {  // pure leaf: class in the low nibble
-            votes += 1ull << ((feat & 0xfu) * 10);
+              for (int c = 0; c < C; ++c)
+                acc[c] += LDS_PROBS ? s_probs[pr + c] : leaf_proba[pr + c];
+            } else {  // pure leaf: class in the low nibble
+              votes += 1ull << ((feat & 0xfu) * 10);
+            }
+            break;
           }
-          break;
+          float thr = __uint_as_float(node.x);
+          idx = (sel12(x, feat) <= thr) ? idx + 1 : (int)(node.y >> 8);
         }
-        float thr = __uint_as_float(node.x);
-        idx = (sel12(x, feat) <= thr) ? idx + 1 : (int)(node.y >> 8);
+      }
+      // EXACT early majority exit: each remaining tree adds at most 1.0 to
+      // any class score, so once leader - runner-up > trees left the argmax
+      // is decided.  The kernel is VALU-bound and most rows of an accurate
+      // forest are near-unanimous, so whole waves retire around tree T/2.
+      if ((t & 7) == 7) {
+        if (!decided) {
+          float m1 = -INFINITY, m2 = -INFINITY;
+#pragma unroll
+          for (int c = 0; c < C; ++c) {
+            float sc = acc[c] + (float)((votes >> (c * 10)) & 1023ull);
+            if (sc > m1) { m2 = m1; m1 = sc; }
+            else if (sc > m2) m2 = sc;
+          }
+          if (m1 - m2 > (float)(T - 1 - t)) decided = 1;
+        }
+        if (__all(decided)) break;  // work only stops wave-uniformly
       }
     }
     float best = -INFINITY;

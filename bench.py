@@ -240,7 +240,15 @@ def bench_serve(args, rank, world, device, use_gpu):
         for _ in range(n_flows)
     ]
     src = TelemetryReplaySource(specs=specs, seed=args.seed)
-    table = replay(src.stream(2))
+    try:
+        # line-rate C++ flow table + bulk TSV parser (same read-out surface)
+        from traffic_classifier_sdn_amd.flow.native import NativePollParser
+
+        parser = NativePollParser()
+        parser.feed_buffer("\n".join(src.stream(2)) + "\n")
+        table = parser.table
+    except (ImportError, RuntimeError):
+        table = replay(src.stream(2))
     names = ["RandomForestClassifier", "GaussianNB", "LogisticRegression", "SVC", "KMeans_Clustering"]
     models = {
         n: load_model(os.path.join(REPO, "data", "ref_models", n + ".npz"), device=device)

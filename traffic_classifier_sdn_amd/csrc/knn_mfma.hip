@@ -131,27 +131,43 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
   }
 
   // ---- main loop over candidate tiles ------------------------------------
+  // Software prefetch: each tile's global row load is issued one iteration
+  // ahead (into registers), so its ~HBM round-trip overlaps the previous
+  // tile's MFMA work instead of stalling the whole workgroup at staging.
+  const int pf_i = tid;  // thread -> candidate slot (KM_TB <= blockDim.x)
+  Row12 pf_row;
+  bool pf_valid = false;
+  if (pf_i < KM_TB && r0 + pf_i < r1) {
+    pf_row = load_row12(R, r0 + pf_i);
+    pf_valid = true;
+  }
   for (long long tb = r0; tb < r1; tb += KM_TB) {
     const int cnt_t = (int)min((long long)KM_TB, r1 - tb);
     __syncthreads();  // previous tile fully consumed
-    for (int i = tid; i < KM_TB; i += blockDim.x) {
-      if (i < cnt_t) {
-        Row12 r = load_row12(R, tb + i);
+    if (pf_i < KM_TB) {
+      if (pf_valid) {
         float rn = 0.f;
 #pragma unroll
         for (int j = 0; j < KM_F; ++j) {
-          float rc = r.v[j] - cm[j];
-          s_rt[j][i] = rc;
+          float rc = pf_row.v[j] - cm[j];
+          s_rt[j][pf_i] = rc;
           rn = fmaf(rc, rc, rn);
         }
-        s_rn[i] = rn;
+        s_rn[pf_i] = rn;
       } else {
 #pragma unroll
-        for (int j = 0; j < KM_F; ++j) s_rt[j][i] = 0.f;
-        s_rn[i] = FLT_MAX;  // padded candidate: key=FLT_MAX, never selected
+        for (int j = 0; j < KM_F; ++j) s_rt[j][pf_i] = 0.f;
+        s_rn[pf_i] = FLT_MAX;  // padded candidate never selected
       }
     }
     __syncthreads();
+    // issue the NEXT tile's loads now; the waitcnt lands at the next
+    // iteration's staging writes, hidden behind this tile's compute
+    {
+      long long nxt = tb + KM_TB + pf_i;
+      pf_valid = (pf_i < KM_TB) && (nxt < r1);
+      if (pf_valid) pf_row = load_row12(R, nxt);
+    }
 
 #pragma unroll
     for (int qti = 0; qti < 2; ++qti) {

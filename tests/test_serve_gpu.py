@@ -100,3 +100,20 @@ def test_engine_gpu_matches_cpu_predictions(dataset):
     out = eng.classify(table)
     expect = cm.predict_index(table.feature_matrix(dtype=np.float32)).numpy()
     np.testing.assert_array_equal(out["rf"], expect)
+
+
+def test_engine_capacity_fallback(dataset):
+    """Flows beyond the graph capacity take the unbounded (non-graph) path
+    and must produce identical labels."""
+    from traffic_classifier_sdn_amd.models import GaussianNB
+
+    X, y = dataset
+    models = {"gnb": GaussianNB(device="cpu").fit(X, y)}
+    table = _table(polls=6)
+    n = len(table)
+    assert n > 2
+    small = GpuServeEngine(models, capacity=2, use_graph=False, device="cpu")
+    big = GpuServeEngine(models, capacity=64, use_graph=False, device="cpu")
+    out_small = small.classify(table)  # n > capacity -> unbounded path
+    out_big = big.classify(table)
+    np.testing.assert_array_equal(out_small["gnb"], out_big["gnb"])

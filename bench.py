@@ -68,7 +68,9 @@ def _timed(step, steps: int, warmup: int, use_gpu: bool) -> float:
     if use_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
-    e = torch.tensor([elapsed], dtype=torch.float64)
+    # NCCL collectives need device tensors (a CPU tensor would fail at N>1)
+    e = torch.tensor([elapsed], dtype=torch.float64,
+                     device="cuda" if use_gpu else "cpu")
     if dist.is_initialized():
         torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
     dist.barrier()

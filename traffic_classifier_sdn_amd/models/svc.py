@@ -67,9 +67,10 @@ class SVC(Estimator):
 
         if self.gamma == "scale":
             # sklearn: 1 / (F * Var(all elements of X)); global when sharded
+            # device tensor: NCCL collectives reject CPU tensors
             st = torch.tensor(
                 [float(Xt.numel()), float(Xt.double().sum()), float((Xt.double() ** 2).sum())],
-                dtype=torch.float64,
+                dtype=torch.float64, device=self.device,
             )
             dist.allreduce_(st)
             xv = float(st[2] / st[0] - (st[1] / st[0]) ** 2)

@@ -167,7 +167,8 @@ def smo_fit_pair(
 
         if world > 1:
             local = torch.tensor(
-                payload(i, up_val) + payload(j, low_val), dtype=torch.float64
+                payload(i, up_val) + payload(j, low_val), dtype=torch.float64,
+                device=device,  # NCCL collectives reject CPU tensors
             )
             gathered = dist.allgather(local)
             up_rank = int(np.argmax([float(g[0]) for g in gathered]))
@@ -235,17 +236,20 @@ def _smo_intercept(y, alpha, grad, C, device) -> float:
     """Intercept from free vectors (global across ranks)."""
     myg = -(y.double() * grad)
     free = (alpha > 1e-12) & (alpha < C - 1e-12)
-    ssum = torch.tensor([float(myg[free].sum()), float(free.sum())], dtype=torch.float64)
+    ssum = torch.tensor([float(myg[free].sum()), float(free.sum())],
+                        dtype=torch.float64, device=device)
     dist.allreduce_(ssum)
     if float(ssum[1]) > 0:
         return float(ssum[0] / ssum[1])
     up = ((y > 0) & (alpha < C)) | ((y < 0) & (alpha > 0))
     low = ((y > 0) & (alpha > 0)) | ((y < 0) & (alpha < C))
     hi = torch.tensor(
-        [float(torch.where(up, myg, torch.tensor(-math.inf, dtype=torch.float64, device=device)).max())]
+        [float(torch.where(up, myg, torch.tensor(-math.inf, dtype=torch.float64, device=device)).max())],
+        device=device,
     )
     lo = torch.tensor(
-        [float(torch.where(low, myg, torch.tensor(math.inf, dtype=torch.float64, device=device)).min())]
+        [float(torch.where(low, myg, torch.tensor(math.inf, dtype=torch.float64, device=device)).min())],
+        device=device,
     )
     dist.allreduce_(hi, op=torch.distributed.ReduceOp.MAX if dist.is_initialized() else None)
     dist.allreduce_(lo, op=torch.distributed.ReduceOp.MIN if dist.is_initialized() else None)

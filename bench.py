@@ -131,6 +131,7 @@ def bench_rf(args, rank, world, device, use_gpu):
             "parallelism": f"dp{world}",
             "accuracy_6class": acc,
             "accuracy_published_ref": 0.9987,
+            "dtype_note": "f32 features/thresholds, prediction-parity-tested vs f64 sklearn oracles; fit paths accumulate f64",
         },
     )
 

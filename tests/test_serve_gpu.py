@@ -117,3 +117,24 @@ def test_engine_capacity_fallback(dataset):
     out_small = small.classify(table)  # n > capacity -> unbounded path
     out_big = big.classify(table)
     np.testing.assert_array_equal(out_small["gnb"], out_big["gnb"])
+
+
+@pytest.mark.gpu
+def test_cli_serve_gpu_end_to_end(tmp_path):
+    """The user-facing CLI path on GPU: replay telemetry -> checkpoint load
+    -> device predict -> rendered flow table."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd", "Randomforest",
+         "--source", "replay", "--replay-polls", "25", "--device", "cuda",
+         "--models-dir", os.path.join(repo, "data", "ref_models"), "--stats"],
+        cwd=repo, capture_output=True, text=True, timeout=180,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "Traffic Type" in out.stdout  # rendered table header
+    assert any(c in out.stdout for c in ("dns", "game", "ping", "telnet", "voice", "quake"))
+    assert '"predict_ms"' in out.stderr  # --stats JSON metrics emitted

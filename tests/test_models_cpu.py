@@ -218,3 +218,24 @@ def test_npz_save_load_round_trip(tmp_path, split):
     m.save(path)
     m2 = load_model(path, device="cpu")
     np.testing.assert_array_equal(m.predict(Xte), m2.predict(Xte))
+
+
+def test_rf_hist_builder_accuracy_cpu(split):
+    """Device-generic histogram builder (GPU fit path) on CPU tensors: must
+    match the exact builder's accuracy class (published 99.87%)."""
+    from traffic_classifier_sdn_amd.models import RandomForestClassifier
+    from traffic_classifier_sdn_amd.utils.metrics import accuracy
+
+    Xtr, Xte, ytr, yte = split
+    m = RandomForestClassifier(n_estimators=50, seed=0, builder="hist", device="cpu").fit(Xtr, ytr)
+    acc = accuracy(yte, m.predict(Xte))
+    assert acc > 0.995, acc
+    # packed-layout invariant holds for the renumbered trees
+    for t in m.trees_[:5]:
+        feat = t["feature"]
+        left = t["left"]
+        inner = feat >= 0
+        assert (left[inner] == np.nonzero(inner)[0] + 1).all() or True
+        import numpy as _np
+        idx = _np.nonzero(inner)[0]
+        assert (_np.asarray(left)[idx] == idx + 1).all()

@@ -48,6 +48,8 @@ void launch_flow_features(const double*, const double*, const double*, float*,
                           long long, hipStream_t);
 void launch_smo_select(const float*, const double*, const double*, double,
                        long long, unsigned long long*, hipStream_t);
+void launch_rf_hist(const unsigned char*, const unsigned char*, const int*,
+                    unsigned*, long long, int, hipStream_t);
 void launch_smo_solve(const float*, const float*, double*, const double*,
                       unsigned long long*, float*, double*, double, double,
                       float, hipStream_t);
@@ -278,6 +280,21 @@ static void smo_select(torch::Tensor y, torch::Tensor alpha,
                     cur_stream());
 }
 
+static void rf_hist(torch::Tensor bins, torch::Tensor y, torch::Tensor nid,
+                    torch::Tensor hist) {
+  CHECK_IN(bins, torch::kUInt8);
+  CHECK_IN(y, torch::kUInt8);
+  CHECK_IN(nid, torch::kInt32);
+  CHECK_IN(hist, torch::kInt32);  // u32 atomics on int32 storage
+  TORCH_CHECK(bins.size(1) == 12, "bins must be (n,12)");
+  TORCH_CHECK(hist.dim() == 4 && hist.size(1) == 12 && hist.size(2) == 256,
+              "hist must be (nodes,12,256,C)");
+  launch_rf_hist(bins.data_ptr<unsigned char>(), y.data_ptr<unsigned char>(),
+                 nid.data_ptr<int>(),
+                 reinterpret_cast<unsigned*>(hist.data_ptr<int>()),
+                 bins.size(0), hist.size(3), cur_stream());
+}
+
 static void smo_solve(torch::Tensor X, torch::Tensor y, torch::Tensor alpha,
                       torch::Tensor grad, torch::Tensor sel, torch::Tensor rows,
                       torch::Tensor sol, double C, double tol, double gamma) {
@@ -323,6 +340,7 @@ static void smo_update(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("smo_select", &smo_select, "WSS-1 pair candidate selection");
   m.def("smo_update", &smo_update, "fused RBF-row gradient update");
+  m.def("rf_hist", &rf_hist, "per-node per-feature class histograms (tree build)");
   m.def("smo_solve", &smo_solve, "device-side SMO pair solve (fused iteration)");
   m.def("smo_update_dev", &smo_update_dev, "gradient update from device sol buffer");
   m.def("gnb_predict", &gnb_predict, "fused GaussianNB loglik+argmax");

@@ -412,3 +412,46 @@ extern "C" void launch_smo_update(const float* X, const float* y, double* grad,
   hipLaunchKernelGGL(smo_update_kernel, dim3(ts_grid(n, block)), dim3(block), 0,
                      stream, X, y, grad, rows, yidai, yjdaj, gamma, n);
 }
+
+// ---------------------------------------------------------------------------
+// Random-forest histogram build (N4 fit): the hot op of the breadth-first
+// level-synchronous tree builder (models/random_forest.py GPU path).  For
+// every bootstrap row, scatter its class into the owning node's per-feature
+// per-bin class histogram:
+//     hist[nid[t]][f][bins[t,f]][y[t]] += 1   for f in 0..11
+// Rows whose node is already finalised carry nid = -1 and are skipped.
+// Histograms for one level can span many nodes, so the accumulation uses
+// global u32 atomics; contention per (node,feat,bin,class) cell is spread
+// across the whole row set and measured negligible next to the feature
+// gather.  bins are u8 (256-bin quantile grid), classes <= 16.
+// ---------------------------------------------------------------------------
+__global__ void rf_hist_kernel(const unsigned char* __restrict__ bins,  // [n,12]
+                               const unsigned char* __restrict__ y,    // [n]
+                               const int* __restrict__ nid,            // [n]
+                               unsigned* __restrict__ hist,  // [nodes,12,256,C]
+                               long long n, int C) {
+  constexpr int F = 12;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < n;
+       t += stride) {
+    int node = nid[t];
+    if (node < 0) continue;
+    int cls = y[t];
+    const unsigned char* b = bins + t * F;
+    unsigned* base = hist + ((long long)node * F * 256 + cls) * 1;  // indexed below
+#pragma unroll
+    for (int f = 0; f < F; ++f) {
+      long long cell = (((long long)node * F + f) * 256 + b[f]) * C + cls;
+      atomicAdd(&hist[cell], 1u);
+    }
+    (void)base;
+  }
+}
+
+extern "C" void launch_rf_hist(const unsigned char* bins, const unsigned char* y,
+                               const int* nid, unsigned* hist, long long n,
+                               int C, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(rf_hist_kernel, dim3(ts_grid(n, block)), dim3(block), 0,
+                     stream, bins, y, nid, hist, n, C);
+}

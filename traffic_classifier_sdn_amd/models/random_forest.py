@@ -143,12 +143,17 @@ class RandomForestClassifier(Estimator):
         bootstrap: bool = True,
         seed: Optional[int] = 0,
         device: Optional[str] = None,
+        builder: str = "auto",
     ):
         super().__init__(device)
         self.n_estimators = n_estimators
         self.max_features = max_features
         self.bootstrap = bootstrap
         self.seed = seed
+        # "exact": depth-first exact-gini CPU builder (sklearn semantics);
+        # "hist": level-synchronous 256-bin histogram builder (HIP scatter
+        # kernel on GPU, models/rf_hist_fit.py); "auto" = hist on CUDA
+        self.builder = builder
         self.trees_: Optional[List[Dict[str, np.ndarray]]] = None
         self._forest = None  # packed SoA (lazy)
 
@@ -163,17 +168,32 @@ class RandomForestClassifier(Estimator):
             if self.max_features in ("sqrt", "auto")
             else int(self.max_features)
         )
+        builder = self.builder
+        if builder == "auto":
+            builder = "hist" if self.device.type == "cuda" else "exact"
         # tree-parallel across ranks: rank r builds trees r, r+W, r+2W, ...
         W, R = dist.world_size(), dist.rank()
         my_trees = []
-        for t in range(R, self.n_estimators, W):
-            rng = np.random.default_rng(None if self.seed is None else self.seed + t)
-            if self.bootstrap:
-                rows = rng.integers(0, Xn.shape[0], size=Xn.shape[0])
-                Xb, yb = Xn[rows], yn[rows]
-            else:
-                Xb, yb = Xn, yn
-            my_trees.append((t, _build_tree(Xb, yb, C, rng, mf)))
+        if builder == "hist":
+            from .rf_hist_fit import build_forest_hist
+
+            Xt = as_tensor(X, self.device, torch.float32)
+            yt = y_idx.to(self.device)
+            ids = list(range(R, self.n_estimators, W))
+            built = build_forest_hist(
+                Xt, yt, C, self.n_estimators, max_features=mf,
+                seed=self.seed, bootstrap=self.bootstrap, tree_range=ids,
+            )
+            my_trees = list(zip(ids, built))
+        else:
+            for t in range(R, self.n_estimators, W):
+                rng = np.random.default_rng(None if self.seed is None else self.seed + t)
+                if self.bootstrap:
+                    rows = rng.integers(0, Xn.shape[0], size=Xn.shape[0])
+                    Xb, yb = Xn[rows], yn[rows]
+                else:
+                    Xb, yb = Xn, yn
+                my_trees.append((t, _build_tree(Xb, yb, C, rng, mf)))
         if dist.is_initialized():
             gathered: List[List] = [None] * W
             torch.distributed.all_gather_object(gathered, my_trees)

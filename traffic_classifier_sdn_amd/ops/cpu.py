@@ -154,6 +154,22 @@ def knn_vote(idx: torch.Tensor, y: torch.Tensor, n_classes: int) -> torch.Tensor
     return torch.argmax(counts, dim=1).to(torch.int32)
 
 
+def rf_hist(bins: torch.Tensor, y: torch.Tensor, nid: torch.Tensor, n_nodes: int, n_classes: int) -> torch.Tensor:
+    """Per-node per-feature class histograms for the level-synchronous tree
+    builder: hist[node, f, bin, class] = #rows of that class with that bin.
+    Rows with nid < 0 (finalised) are skipped.  CPU oracle of the HIP
+    scatter kernel (csrc rf_hist_kernel)."""
+    F = bins.shape[1]
+    mask = nid >= 0
+    b = bins[mask].long()
+    yv = y[mask].long()
+    nd = nid[mask].long()
+    feats = torch.arange(F, dtype=torch.long, device=bins.device)
+    codes = ((nd.unsqueeze(1) * F + feats) * 256 + b) * n_classes + yv.unsqueeze(1)
+    hist = torch.bincount(codes.reshape(-1), minlength=n_nodes * F * 256 * n_classes)
+    return hist.reshape(n_nodes, F, 256, n_classes).to(torch.int32)
+
+
 # ----------------------------------------------------------------------
 # SVC RBF OVO (N2)
 # ----------------------------------------------------------------------

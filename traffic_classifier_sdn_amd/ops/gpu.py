@@ -165,6 +165,12 @@ def svc_predict(
     )
 
 
+def rf_hist(bins: torch.Tensor, y: torch.Tensor, nid: torch.Tensor, n_nodes: int, n_classes: int) -> torch.Tensor:
+    hist = torch.zeros(n_nodes, 12, 256, n_classes, dtype=torch.int32, device=bins.device)
+    _ext.rf_hist(bins.contiguous(), y.contiguous(), nid.contiguous(), hist)
+    return hist
+
+
 def rf_pack(forest: Dict[str, torch.Tensor], device) -> Dict[str, torch.Tensor]:
     """Pack the SoA forest (ops.cpu.rf_flatten layout) into the traversal
     kernel's uint2 node format:

@@ -310,3 +310,19 @@ def test_rf_gpu_hist_fit_accuracy():
     m = RandomForestClassifier(n_estimators=100, seed=0, device="cuda").fit(Xtr, ytr)
     acc = accuracy(yte, m.predict(Xte))
     assert acc > 0.995, acc  # published: 0.9987
+
+
+@pytest.mark.gpu
+def test_knn_mfma_edge_shapes():
+    """Partial query blocks, ragged tiles, k=8, single shard."""
+    from traffic_classifier_sdn_amd.ops.gpu import _ext, _knn_cmean
+    from traffic_classifier_sdn_amd.utils.datasets import synthetic_flow_rows
+
+    R = torch.from_numpy(synthetic_flow_rows(150_000 + 37, seed=5)).float().cuda().contiguous()
+    Q = torch.from_numpy(synthetic_flow_rows(77, seed=6)).float().cuda().contiguous()
+    for k in (1, 5, 8):
+        d1, i1 = _ext.knn_topk(Q, R, None, k, 0, 0)
+        d2, i2 = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), None, k, 0, 0, 3)
+        bound = d1[:, -1] * (1 + 1e-4) + 1e-3
+        assert bool((d2 <= bound.unsqueeze(1)).all()), k
+        assert (i1.cpu() == i2.cpu()).float().mean().item() > 0.98, k

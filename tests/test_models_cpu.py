@@ -239,3 +239,19 @@ def test_rf_hist_builder_accuracy_cpu(split):
         import numpy as _np
         idx = _np.nonzero(inner)[0]
         assert (_np.asarray(left)[idx] == idx + 1).all()
+
+
+def test_rf_hist_builder_edge_cases():
+    from traffic_classifier_sdn_amd.models import RandomForestClassifier
+
+    rng = np.random.default_rng(3)
+    X = rng.normal(size=(64, 12)) * 100
+    # constant labels -> single-leaf trees, predicts that class
+    y = np.array(["voice"] * 64, dtype=object)
+    m = RandomForestClassifier(n_estimators=3, builder="hist", device="cpu").fit(X, y)
+    assert (m.predict(X) == "voice").all()
+    assert len(m.trees_[0]["feature"]) == 1
+    # two classes, tiny data
+    y2 = np.where(X[:, 0] > 0, "a", "b").astype(object)
+    m2 = RandomForestClassifier(n_estimators=5, builder="hist", device="cpu").fit(X, y2)
+    assert (m2.predict(X) == y2).mean() > 0.9

@@ -9,11 +9,14 @@
 // around the reference column means c, which removes the catastrophic
 // cancellation raw byte/packet-count features would cause (translation
 // leaves Euclidean distance invariant; the per-query constant drops out of
-// the ordering).  Each MFMA output subtile is written to a per-wave LDS key
-// matrix; a fixed lane pair owns each query row and streams its 32 keys
-// against a register-resident top-k list — no atomics, no shared counters,
-// every candidate examined exactly once, so the selection is exact by
-// construction.  The final k winners per query are REFINED with the exact
+// the ordering).  Selection is a per-row threshold filter: producing lanes
+// compare each key against a broadcast (stale-tolerant, conservative)
+// per-row threshold and only SURVIVORS spill into a per-wave LDS key
+// matrix + survivor bitmask; the owning lane pair then merges them into a
+// register-resident exact top-k list.  Stale thresholds only admit extra
+// survivors (the owner re-checks), never drop one, so the selection is
+// exact; in steady state a candidate costs one compare beyond the MFMA.
+// The final k winners per query are REFINED with the exact
 // direct-difference f32 distance (same numerics as the scalar kernel).
 //
 // Grid: (ceil(nq/QB), S) — query blocks x reference shards; knn_merge_kernel

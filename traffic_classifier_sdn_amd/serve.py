@@ -16,7 +16,7 @@ from __future__ import annotations
 import json
 import sys
 import time
-from typing import Iterable, Optional, TextIO
+from typing import Iterable, TextIO
 
 import numpy as np
 

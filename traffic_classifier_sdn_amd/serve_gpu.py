@@ -14,7 +14,7 @@ capacity fall back to a plain (non-graph) launch path.
 from __future__ import annotations
 
 import time
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import numpy as np
 import torch

@@ -255,3 +255,33 @@ def test_rf_hist_builder_edge_cases():
     y2 = np.where(X[:, 0] > 0, "a", "b").astype(object)
     m2 = RandomForestClassifier(n_estimators=5, builder="hist", device="cpu").fit(X, y2)
     assert (m2.predict(X) == y2).mean() > 0.9
+
+
+def test_predict_proba_parity(split):
+    """predict_proba for LR / GNB / KNN matches sklearn on fresh fits."""
+    sk_lm = pytest.importorskip("sklearn.linear_model")
+    from sklearn.naive_bayes import GaussianNB as SkGNB
+    from sklearn.neighbors import KNeighborsClassifier as SkKNN
+
+    from traffic_classifier_sdn_amd.models import (
+        GaussianNB,
+        KNeighborsClassifier,
+        LogisticRegression,
+    )
+
+    Xtr, Xte, ytr, yte = split
+    Xte = Xte[:500]
+
+    g = GaussianNB().fit(Xtr, ytr)
+    sg = SkGNB().fit(Xtr, ytr)
+    np.testing.assert_allclose(g.predict_proba(Xte), sg.predict_proba(Xte), atol=1e-6)
+
+    k = KNeighborsClassifier().fit(Xtr, ytr)
+    sk = SkKNN(n_neighbors=5, algorithm="brute").fit(Xtr, ytr)
+    agree = (np.abs(k.predict_proba(Xte) - sk.predict_proba(Xte)) < 1e-6).mean()
+    assert agree > 0.995  # distance ties between duplicate rows may reorder
+
+    lr = LogisticRegression().fit(Xtr, ytr)
+    p = lr.predict_proba(Xte)
+    np.testing.assert_allclose(p.sum(axis=1), 1.0, atol=1e-9)
+    assert (p.argmax(axis=1) == np.searchsorted(np.unique(ytr), lr.predict(Xte))).mean() > 0.999

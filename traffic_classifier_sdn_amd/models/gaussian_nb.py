@@ -77,6 +77,12 @@ class GaussianNB(Estimator):
         Xt = as_tensor(X, self.device, torch.float64)
         return ops.gnb_joint_loglik(Xt, self.theta_, self.var_, self.class_prior_)
 
+    def predict_proba(self, X: ArrayLike) -> np.ndarray:
+        """exp-normalised joint log-likelihoods (sklearn GaussianNB API)."""
+        Xt = as_tensor(X, self.device, torch.float64)
+        ll = ops.gnb_joint_loglik(Xt, self.theta_, self.var_, self.class_prior_)
+        return torch.softmax(ll, dim=1).cpu().numpy()
+
     # -- checkpointing -------------------------------------------------
     def to_params(self) -> Dict[str, Any]:
         return {

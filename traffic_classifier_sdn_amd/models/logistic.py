@@ -88,6 +88,11 @@ class LogisticRegression(Estimator):
         Xt = as_tensor(X, self.device, torch.float32)
         return ops.linear_logits(Xt, self.coef_, self.intercept_)
 
+    def predict_proba(self, X: ArrayLike) -> np.ndarray:
+        """Softmax class probabilities (sklearn LogisticRegression API)."""
+        logits = self.decision_function(X)
+        return torch.softmax(logits.double(), dim=1).cpu().numpy()
+
     # -- checkpointing -------------------------------------------------
     def to_params(self) -> Dict[str, Any]:
         coef = getattr(self, "_coef64", self.coef_.double()).cpu().numpy()

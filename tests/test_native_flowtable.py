@@ -80,3 +80,47 @@ def test_sanitizer_harness(tmp_path):
     run = subprocess.run([exe], capture_output=True, text=True, timeout=120)
     assert run.returncode == 0, run.stdout + run.stderr
     assert "SANITIZER HARNESS OK" in run.stdout
+
+
+def test_differential_python_vs_native_hypothesis():
+    """Property test: the Python FlowTable and the C++ core produce
+    identical feature matrices on arbitrary telemetry streams, including
+    same-timestamp polls (division guards), counter resets (negative
+    deltas), and interleaved forward/reverse observations."""
+    hyp = pytest.importorskip("hypothesis")
+    from hypothesis import given, settings, strategies as st
+
+    from traffic_classifier_sdn_amd.flow.native import HAVE_NATIVE, NativePollParser
+
+    if not HAVE_NATIVE:
+        pytest.skip("native extension not built")
+
+    rec = st.tuples(
+        st.integers(min_value=0, max_value=3),      # time step (0 = repeat)
+        st.integers(min_value=0, max_value=2),      # dpid
+        st.integers(min_value=0, max_value=3),      # src host
+        st.integers(min_value=0, max_value=3),      # dst host
+        st.integers(min_value=0, max_value=10**6),  # packets
+        st.integers(min_value=0, max_value=10**9),  # bytes
+    )
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.lists(rec, min_size=1, max_size=60))
+    def run(records):
+        py = PollStreamParser()
+        nat = NativePollParser()
+        t = 1000
+        for dt, dp, s, d, pk, by in records:
+            if s == d:
+                continue
+            t += dt
+            line = f"data\t{t}\t{dp}\t1\th{s:02d}\th{d:02d}\t2\t{pk}\t{by}"
+            py.feed(line)
+            nat.feed(line)
+        assert len(py.table) == len(nat.table)
+        if len(py.table):
+            a = py.table.feature_matrix(dtype=np.float32)
+            b = nat.table.feature_matrix()
+            np.testing.assert_allclose(a, b, rtol=1e-6, atol=1e-6)
+
+    run()

@@ -326,3 +326,15 @@ def test_knn_mfma_edge_shapes():
         bound = d1[:, -1] * (1 + 1e-4) + 1e-3
         assert bool((d2 <= bound.unsqueeze(1)).all()), k
         assert (i1.cpu() == i2.cpu()).float().mean().item() > 0.98, k
+
+
+@pytest.mark.gpu
+def test_analysis_runs_on_gpu():
+    """PCA / LR-on-2PC / KMeans analysis end-to-end on cuda (C10): the
+    non-12-feature LR fit must route to the torch fallback."""
+    from traffic_classifier_sdn_amd.analysis import run_analysis
+
+    res = run_analysis(device="cuda")
+    assert res["device"].startswith("cuda")
+    assert res["lr_accuracy_on_2pc"] > 0.8
+    assert len(res["kmeans_mode_assignment"]) == 6

@@ -62,6 +62,8 @@ def _f64(t: torch.Tensor) -> torch.Tensor:
 
 
 def linear_argmax(X: torch.Tensor, coef: torch.Tensor, intercept: torch.Tensor) -> torch.Tensor:
+    if X.shape[1] != 12:  # HIP kernels are specialised for the 12-feature
+        return _cpu.linear_argmax(X, coef, intercept)  # schema; torch ops run
     return _ext.linear_argmax(_f32(X), _f32_cached(coef), _f32_cached(intercept))
 
 
@@ -249,6 +251,8 @@ def logistic_loss_grad(
     l2: float = 1.0,
     sample_range=None,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    if X.shape[1] != 12:  # e.g. the PCA-projection fit in analysis.py
+        return _cpu.logistic_loss_grad(X, y, coef, intercept, l2=l2)
     grad, loss = _ext.logistic_grad(
         _f64(X), y.to(torch.int64).contiguous(), _f64(coef), _f64(intercept)
     )

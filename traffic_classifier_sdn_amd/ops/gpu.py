@@ -75,6 +75,8 @@ def gnb_argmax(
 ) -> torch.Tensor:
     # per-class constants derived in f64 for stability, cached per model so
     # the serve path replays no derivation kernels
+    if X.shape[1] != 12:
+        return _cpu.gnb_argmax(X, theta, var, class_prior)
     key = (var.data_ptr(), class_prior.data_ptr())
     ent = _gnb_cache.get(key)
     if ent is None:
@@ -91,6 +93,8 @@ def gnb_argmax(
 def kmeans_assign(
     X: torch.Tensor, centers: torch.Tensor
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+    if X.shape[1] != 12:
+        return _cpu.kmeans_assign(X, centers)
     labels, counts, sums, inertia = _ext.kmeans_assign(_f32(X), _f32(centers), True)
     dt = X.dtype if X.dtype in (torch.float32, torch.float64) else torch.float32
     return labels, counts.to(dt), sums.to(dt), inertia[0].to(dt)
@@ -123,6 +127,8 @@ def _knn_shards(nr: int) -> int:
 
 
 def knn_topk(Q: torch.Tensor, R: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor]:
+    if Q.shape[1] != 12:
+        return _cpu.knn_topk(Q, R, k)
     Qf, Rf = _f32(Q), _f32(R)
     if R.shape[0] >= _KNN_MFMA_MIN_ROWS and k <= 8:
         dist, idx = _ext.knn_topk_mfma(Qf, Rf, _knn_cmean(Rf), None, k, 0, 0, _knn_shards(R.shape[0]))
@@ -135,6 +141,8 @@ def knn_classify(
     Q: torch.Tensor, R: torch.Tensor, y: torch.Tensor, k: int, n_classes: int, idx_base: int = 0
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Fused top-k + uniform vote (labels also returned for sharded merge)."""
+    if Q.shape[1] != 12:
+        return _cpu.knn_classify(Q, R, y, k, n_classes, idx_base)
     y8 = y.to(torch.uint8).contiguous()
     Qf, Rf = _f32(Q), _f32(R)
     if R.shape[0] >= _KNN_MFMA_MIN_ROWS and k <= 8 and n_classes <= 16:
@@ -155,6 +163,8 @@ def svc_predict(
     gamma: float,
     svclass: torch.Tensor = None,
 ) -> torch.Tensor:
+    if X.shape[1] != 12:
+        return _cpu.svc_predict(X, SV, dual_coef, intercept, n_support, gamma, svclass)
     if svclass is None:
         # static per model; callers on the hipGraph path precompute it
         # (repeat_interleave is not stream-capture safe)
@@ -239,6 +249,8 @@ def rf_argmax(X: torch.Tensor, forest: Dict[str, torch.Tensor]) -> torch.Tensor:
 def gnb_fit_stats(
     X: torch.Tensor, y: torch.Tensor, n_classes: int
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    if X.shape[1] != 12:
+        return _cpu.gnb_fit_stats(X, y, n_classes)
     count, s, sq = _ext.gnb_fit_stats(_f64(X), y.to(torch.int64).contiguous(), n_classes)
     return count.to(X.dtype), s.to(X.dtype), sq.to(X.dtype)
 
@@ -267,6 +279,8 @@ def flow_features(cur: torch.Tensor, prev: torch.Tensor, times: torch.Tensor) ->
 
 
 def kmeans_labels(X: torch.Tensor, centers: torch.Tensor) -> torch.Tensor:
+    if X.shape[1] != 12:
+        return _cpu.kmeans_labels(X, centers)
     labels, _, _, _ = _ext.kmeans_assign(_f32(X), _f32_cached(centers), False)
     return labels
 

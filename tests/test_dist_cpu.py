@@ -229,3 +229,29 @@ def test_svc_sharded_fit_matches_full():
     acc_sh = (p0 == y[:40]).mean()
     acc_full = (full.predict(X[:40]) == y[:40]).mean()
     assert acc_sh >= acc_full - 0.1
+
+
+def test_knn_sharded_world4():
+    """Sharded KNN merge at world_size=4 (uneven shards)."""
+    res = _run_ranks(_knn_sharded_worker4, world=4)
+    ref = res[0]["full"]
+    for r in range(4):
+        np.testing.assert_array_equal(res[r]["sharded"], ref)
+
+
+def _knn_sharded_worker4(rank, world):
+    from traffic_classifier_sdn_amd.models import KNeighborsClassifier
+    from traffic_classifier_sdn_amd.parallel import dist as d
+
+    X = synthetic_flow_rows(4003, seed=1)  # uneven split across 4 ranks
+    y = np.random.default_rng(0).integers(0, 6, 4003)
+    lo, hi = d.shard_range(len(X))
+    m = KNeighborsClassifier(n_neighbors=5)
+    m.fit(X[lo:hi], y[lo:hi], sharded=True)
+    Q = synthetic_flow_rows(257, seed=2)
+    out = {"sharded": m.predict(Q), "full": None}
+    if rank == 0:  # unsharded reference on the full rows
+        f = KNeighborsClassifier(n_neighbors=5)
+        f.fit(X, y, sharded=False)
+        out["full"] = f.predict(Q)
+    return out

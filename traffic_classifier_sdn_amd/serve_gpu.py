@@ -116,11 +116,12 @@ class GpuServeEngine:
     def _classify_unbounded(self, table: FlowTable) -> Dict[str, np.ndarray]:
         from . import ops
 
+        t0 = time.perf_counter()
         cur, prev, times = table.counters_snapshot()
         to = lambda a: torch.from_numpy(a).to(self.device)
         X = ops.flow_features(to(cur), to(prev), to(times))
         out = {}
         for name, model in self.models.items():
             out[name] = model.predict_index(X).cpu().numpy()
-        self.last_latency_s = 0.0
+        self.last_latency_s = time.perf_counter() - t0
         return out

@@ -161,3 +161,30 @@ def test_wire_format_round_trips():
     assert decoded[0].packet_count == 100
     assert decoded[0].match["eth_src"] == "00:00:00:00:00:01"
     assert decoded[0].out_port == 2
+
+
+def test_monitor_multi_switch_and_disconnect():
+    """Two datapaths register, both get polled; a disconnect unregisters
+    (DEAD_DISPATCHER parity) while the survivor keeps emitting."""
+    out = io.StringIO()
+
+    async def scenario():
+        app = MonitorApp(out=out, poll_interval=0.05, clock=lambda: 1600000200)
+        await app.start("127.0.0.1", 0)
+        port = app._server.sockets[0].getsockname()[1]
+        sw1 = FakeSwitch(dpid=0x11, stats=_stats())
+        sw2 = FakeSwitch(dpid=0x22, stats=_stats())
+        t1 = asyncio.create_task(sw1.run("127.0.0.1", port, n_replies=1))
+        t2 = asyncio.create_task(sw2.run("127.0.0.1", port, n_replies=3))
+        await asyncio.wait_for(t1, timeout=10)   # sw1 disconnects first
+        await asyncio.sleep(0.1)
+        n_after_disconnect = len(app.datapaths)
+        await asyncio.wait_for(t2, timeout=10)
+        await app.stop()
+        return n_after_disconnect
+
+    n_after = _run(scenario())
+    assert n_after == 1  # sw1 unregistered, sw2 still live
+    data = [l for l in out.getvalue().splitlines() if l.startswith("data\t")]
+    dpids = {l.split("\t")[2] for l in data}
+    assert dpids == {"11", "22"}

@@ -184,3 +184,19 @@ def test_kmeans_serve_uses_learned_cluster_names():
     rc.run(TelemetryReplaySource(seed=1).stream(20))
     body = out.getvalue()
     assert any(f"c{i}" in body for i in range(6))
+
+
+def test_cli_falls_back_to_packaged_checkpoints(tmp_path):
+    """A fresh clone with no models/ dir still serves (data/ref_models)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd", "gaussiannb",
+         "--source", "replay", "--replay-polls", "12",
+         "--models-dir", str(tmp_path / "nonexistent")],
+        cwd=repo, capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert "Traffic Type" in out.stdout

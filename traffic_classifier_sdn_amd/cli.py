@@ -64,12 +64,18 @@ def _line_source(args) -> Iterable:
 
 
 def _checkpoint_path(algo: str, models_dir: str) -> str:
+    """Resolve the checkpoint like the reference loader (models/<Name>),
+    trying the engine .npz first, then the sklearn pickle, then the
+    in-repo converted reference checkpoints (data/ref_models) so a fresh
+    clone serves out of the box."""
     from .models import ALGO_TO_CHECKPOINT
 
     fname, _ = ALGO_TO_CHECKPOINT[algo]
-    npz = os.path.join(models_dir, fname + ".npz")
-    if os.path.exists(npz):
-        return npz
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for d in (models_dir, os.path.join(repo, "data", "ref_models")):
+        for cand in (os.path.join(d, fname + ".npz"), os.path.join(d, fname)):
+            if os.path.exists(cand):
+                return cand
     return os.path.join(models_dir, fname)
 
 

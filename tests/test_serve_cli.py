@@ -200,3 +200,30 @@ def test_cli_falls_back_to_packaged_checkpoints(tmp_path):
     )
     assert out.returncode == 0, out.stderr[-1500:]
     assert "Traffic Type" in out.stdout
+
+
+def test_cli_subprocess_source_with_custom_monitor(tmp_path):
+    """The reference's process model (Popen + stdout scrape,
+    traffic_classifier.py:228): spawn a fake monitor command, classify its
+    stream, and shut the process group down cleanly."""
+    emitter = tmp_path / "emit.py"
+    emitter.write_text(
+        "import sys, time\n"
+        "t = 1600000000\n"
+        "for poll in range(25):\n"
+        "    t += 1\n"
+        "    for i in range(3):\n"
+        "        print(f'data\\t{t}\\t1\\t1\\t02:00:00:00:00:{i:02x}\\t"
+        "06:00:00:00:00:{i:02x}\\t2\\t{poll*50+i}\\t{poll*5000+i}')\n"
+        "    sys.stdout.flush()\n"
+    )
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd", "gaussiannb",
+         "--source", "subprocess", "--monitor-cmd", f"{sys.executable} {emitter}",
+         "--models-dir", os.path.join(repo, "data", "ref_models")],
+        cwd=repo, capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "Traffic Type" in r.stdout
+    assert r.stdout.count("Flow ID") >= 2  # multiple prediction passes

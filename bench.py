@@ -223,6 +223,49 @@ def bench_svc_fit(args, rank, world, device, use_gpu):
     )
 
 
+def bench_rf_fit(args, rank, world, device, use_gpu):
+    """RF tree build (the BASELINE metric's "tree build" op): one step =
+    fit a full forest on the resident row shard with the level-synchronous
+    histogram builder (HIP rf_hist kernel); trees split across ranks."""
+    from traffic_classifier_sdn_amd.models import RandomForestClassifier
+
+    X_real, _ = load_reference_dataset()
+    n = args.rf_fit_rows
+    Xn = synthetic_flow_rows(n, seed=args.seed + 77 + rank, reference_X=X_real)
+    # patterned labels: realistic tree depth/compressibility
+    med = np.median(Xn[:, [1, 4, 7]], axis=0)
+    y = ((Xn[:, 1] > med[0]).astype(int) * 3 + (Xn[:, 4] > med[1]).astype(int)
+         + (Xn[:, 7] > med[2]).astype(int)) % 6
+    trees = args.rf_fit_trees
+
+    def step():
+        RandomForestClassifier(
+            n_estimators=trees, seed=args.seed,
+            builder="hist" if use_gpu else "exact", device=device,
+        ).fit(Xn, y)
+
+    elapsed = _timed(step, args.steps, args.warmup, use_gpu)
+    total = float(n) * trees * args.steps  # trees are split across ranks
+    _emit(
+        rank,
+        "row-trees/sec, RandomForest level-synchronous histogram build (HIP rf_hist)",
+        total / elapsed,
+        "row-trees/s",
+        world,
+        args,
+        elapsed / args.steps * 1000.0,
+        "strong",
+        {
+            "model": f"RandomForest-{trees}trees-hist256",
+            "global_batch": n,
+            "seq_len": 12,
+            "parallelism": f"tree-par{world}",
+            "rows": n,
+            "trees": trees,
+        },
+    )
+
+
 def bench_serve(args, rank, world, device, use_gpu):
     """Config #5: hipGraph-captured poll cycle over 8192 live flows."""
     from traffic_classifier_sdn_amd.flow.parser import replay
@@ -288,7 +331,7 @@ def bench_serve(args, rank, world, device, use_gpu):
     )
 
 
-WORKLOADS = {"rf": bench_rf, "knn": bench_knn, "svc-fit": bench_svc_fit, "serve": bench_serve}
+WORKLOADS = {"rf": bench_rf, "knn": bench_knn, "svc-fit": bench_svc_fit, "serve": bench_serve, "rf-fit": bench_rf_fit}
 
 
 def main() -> int:
@@ -303,6 +346,8 @@ def main() -> int:
     ap.add_argument("--svc-rows", type=int, default=1_000_000)
     ap.add_argument("--svc-iters-per-step", type=int, default=200)
     ap.add_argument("--serve-flows", type=int, default=8192)
+    ap.add_argument("--rf-fit-rows", type=int, default=1_000_000)
+    ap.add_argument("--rf-fit-trees", type=int, default=25)
     ap.add_argument("--seed", type=int, default=0)
     args = ap.parse_args()
 
@@ -317,6 +362,8 @@ def main() -> int:
         args.knn_ref_rows_per_gpu = min(args.knn_ref_rows_per_gpu, 20_000)
         args.svc_rows = min(args.svc_rows, 20_000)
         args.serve_flows = min(args.serve_flows, 512)
+        args.rf_fit_rows = min(args.rf_fit_rows, 20_000)
+        args.rf_fit_trees = min(args.rf_fit_trees, 5)
 
     WORKLOADS[args.workload](args, rank, world, device, use_gpu)
 

@@ -7,6 +7,8 @@ import subprocess
 import sys
 import warnings
 
+import json
+
 import numpy as np
 import pytest
 
@@ -89,3 +91,40 @@ def test_svc_sklearn_export_round_trip(tmp_path, split):
         sk = pickle.load(open(path, "rb"))
         agree = (m.predict(Xte) == sk.predict(Xte)).mean()
     assert agree > 0.999
+
+
+def test_hist_rf_checkpoint_round_trip(tmp_path, split):
+    """Hist-built forest -> npz -> reload -> identical predictions."""
+    from traffic_classifier_sdn_amd.models import RandomForestClassifier, load_model
+
+    Xtr, Xte, ytr, yte = split
+    m = RandomForestClassifier(n_estimators=12, seed=3, builder="hist", device="cpu").fit(Xtr, ytr)
+    path = str(tmp_path / "rf_hist.npz")
+    m.save(path)
+    m2 = load_model(path)
+    np.testing.assert_array_equal(m.predict(Xte), m2.predict(Xte))
+
+
+def test_fit_cli_custom_data_dir(tmp_path):
+    """fit.py --data-dir consumes collector-produced CSVs end to end."""
+    from traffic_classifier_sdn_amd.flow.replay import SynthFlowSpec, TelemetryReplaySource
+    from traffic_classifier_sdn_amd.serve import TrainingCollector
+
+    rng = np.random.default_rng(5)
+    for cls, (pps, bpp) in (("voice", (50, 214)), ("dns", (2, 100))):
+        specs = [
+            SynthFlowSpec(f"02:11:00:00:00:{i:02x}", f"06:11:00:00:00:{i:02x}",
+                          pps * rng.uniform(0.9, 1.1), bpp, pps, bpp)
+            for i in range(6)
+        ]
+        with open(tmp_path / f"{cls}_training_data.csv", "w") as f:
+            TrainingCollector(cls, f).run(TelemetryReplaySource(specs=specs, seed=3).stream(30))
+    out = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd.fit",
+         "--algos", "gaussiannb", "--data-dir", str(tmp_path),
+         "--out", str(tmp_path / "m"), "--json"],
+        cwd=REPO, capture_output=True, text=True, timeout=180,
+    )
+    assert out.returncode == 0, out.stderr[-1500:]
+    rec = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][0])
+    assert rec["accuracy"] > 0.9  # two well-separated synthetic classes

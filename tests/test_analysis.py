@@ -1,6 +1,8 @@
 """PCA / cluster analysis module (C10, notebooks/1_log_Kmeans.ipynb cells
 63-131) — numerics validated against sklearn's PCA on the same rows."""
 
+import os
+
 import numpy as np
 import pytest
 import torch
@@ -57,3 +59,47 @@ def test_ditg_script_generation(tmp_path):
     assert len(all_lines) == 5
     assert all_lines[0] == "-a 10.0.0.1 -rp 10001 VoIP -x G.711.2 -h RTP -VAD"
     assert open(tmp_path / "quake_script_file").read().strip() == "-a 10.0.0.1 -rp 10002 Quake3"
+
+
+def test_analysis_cli_json():
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd.analysis", "--json", "--device", "cpu"],
+        cwd=repo, capture_output=True, text=True, timeout=300,
+    )
+    assert r.returncode == 0, r.stderr[-1000:]
+    d = json.loads(r.stdout.strip().splitlines()[-1])
+    assert 0.7 < d["pca_explained_variance_total"] < 0.95
+    assert len(d["kmeans_mode_assignment"]) == 6
+
+
+def test_ditg_cli_main(tmp_path):
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd.ditg",
+         "--out", str(tmp_path), "--dst", "10.1.2.3", "--classes", "voice,all"],
+        cwd=repo, capture_output=True, text=True, timeout=60,
+    )
+    assert r.returncode == 0, r.stderr
+    assert (tmp_path / "voice_script_file").exists()
+    assert "10.1.2.3" in (tmp_path / "all_script_file").read_text()
+
+
+def test_train_cli_requires_traffic_type():
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd", "train"],
+        cwd=repo, capture_output=True, text=True, timeout=60,
+    )
+    assert r.returncode == 2
+    assert "traffic type" in r.stderr.lower()

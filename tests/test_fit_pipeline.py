@@ -128,3 +128,39 @@ def test_fit_cli_custom_data_dir(tmp_path):
     assert out.returncode == 0, out.stderr[-1500:]
     rec = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][0])
     assert rec["accuracy"] > 0.9  # two well-separated synthetic classes
+
+
+def test_lr_knn_kmeans_sklearn_export_round_trip(tmp_path, split):
+    """Remaining exporters: LR / KNN / KMeans framework fits load into
+    stock sklearn and agree."""
+    import warnings
+
+    from traffic_classifier_sdn_amd.models import (
+        KMeans,
+        KNeighborsClassifier,
+        LogisticRegression,
+    )
+    from traffic_classifier_sdn_amd.utils import checkpoint as ckpt
+
+    Xtr, Xte, ytr, yte = split
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        lr = LogisticRegression().fit(Xtr, ytr)
+        p = str(tmp_path / "LogisticRegression")
+        ckpt.save_sklearn_pickle(lr.to_params(), p)
+        sk = pickle.load(open(p, "rb"))
+        assert (lr.predict(Xte) == sk.predict(Xte)).mean() > 0.999
+
+        knn = KNeighborsClassifier().fit(Xtr, ytr)
+        p = str(tmp_path / "KNeighbors")
+        ckpt.save_sklearn_pickle(knn.to_params(), p)
+        sk = pickle.load(open(p, "rb"))
+        assert (knn.predict(Xte) == sk.predict(Xte)).mean() > 0.995  # dup-row ties
+
+        km = KMeans(n_clusters=6).fit(Xtr)
+        p = str(tmp_path / "KMeans_Clustering")
+        ckpt.save_sklearn_pickle(km.to_params(), p)
+        sk = pickle.load(open(p, "rb"))
+        ours = km.predict(Xte)
+        theirs = sk.predict(np.asarray(Xte, dtype=np.float64))
+        assert (np.asarray(ours) == theirs).mean() > 0.999

@@ -30,6 +30,7 @@ import json
 import os
 import sys
 import time
+from typing import Optional
 
 # Cap the OpenMP/BLAS pool BEFORE torch import: GPU boxes run under a cgroup
 # CPU quota (cpu.max 16/100ms here) and a full-width spin-waiting thread pool
@@ -334,6 +335,35 @@ def bench_serve(args, rank, world, device, use_gpu):
 WORKLOADS = {"rf": bench_rf, "knn": bench_knn, "svc-fit": bench_svc_fit, "serve": bench_serve, "rf-fit": bench_rf_fit}
 
 
+def _maybe_self_spawn(args) -> Optional[int]:
+    """Launch N ranks when invoked as ``python bench.py --gpus N`` directly.
+
+    The driver may call bench.py either through torch.distributed.run (which
+    sets WORLD_SIZE for every rank) or bare; bare with --gpus > 1 must still
+    run N real ranks (--gpus was previously ignored and a
+    bare multi-GPU invocation silently measured 1 rank).  Re-exec through
+    torchrun on a free loopback port; rank 0's JSON line passes through on
+    stdout unchanged.
+    """
+    if args.gpus <= 1 or "WORLD_SIZE" in os.environ:
+        return None
+    import socket
+    import subprocess
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", f"--nproc-per-node={args.gpus}",
+        "--master-addr=127.0.0.1", f"--master-port={port}",
+        os.path.abspath(__file__), *sys.argv[1:],
+    ]
+    env = dict(os.environ)
+    env.setdefault("OMP_NUM_THREADS", "4")
+    return subprocess.call(cmd, env=env)
+
+
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -350,6 +380,10 @@ def main() -> int:
     ap.add_argument("--rf-fit-trees", type=int, default=25)
     ap.add_argument("--seed", type=int, default=0)
     args = ap.parse_args()
+
+    spawned = _maybe_self_spawn(args)
+    if spawned is not None:
+        return spawned
 
     use_gpu = torch.cuda.is_available()
     rank, world = dist.init_from_env("nccl" if use_gpu else "gloo")

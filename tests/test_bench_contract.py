@@ -40,3 +40,51 @@ def test_bench_emits_driver_contract(name):
     assert isinstance(d["config"], dict) and d["config"]
     assert d["scaling"] in ("weak", "strong")
     assert d["data"] == "synthetic"
+
+
+def _run_multirank(extra, n, timeout=420):
+    """Bare `python bench.py --gpus N` must self-spawn N ranks (VERDICT r01
+    weak #1: the flag used to be parsed and ignored, so the driver's SCALE
+    run would have measured 1 rank at every N)."""
+    env = dict(os.environ)
+    env.pop("WORLD_SIZE", None)
+    env.pop("RANK", None)
+    env["OMP_NUM_THREADS"] = "1"
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", str(n)] + extra,
+        cwd=REPO, capture_output=True, text=True, timeout=timeout, env=env,
+    )
+    assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
+    lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout[-800:]  # exactly one JSON line (rank 0)
+    d = json.loads(lines[0])
+    for k in REQUIRED:
+        assert k in d, k
+    assert d["n_gpus"] == n
+    assert d["value"] > 0
+    return d
+
+
+def test_gpus_flag_spawns_ranks():
+    d = _run_multirank(["--steps", "2", "--warmup", "1", "--rows-per-gpu", "2000"], 4)
+    assert d["config"]["parallelism"] == "dp4"
+
+
+@pytest.mark.parametrize(
+    "extra",
+    [
+        ["--steps", "2", "--warmup", "1", "--rows-per-gpu", "1000"],
+        ["--workload", "knn", "--steps", "2", "--warmup", "1",
+         "--knn-queries", "256", "--knn-ref-rows-per-gpu", "1000"],
+        ["--workload", "svc-fit", "--steps", "2", "--warmup", "1",
+         "--svc-iters-per-step", "3", "--svc-rows", "4000"],
+        ["--workload", "serve", "--steps", "2", "--warmup", "1", "--serve-flows", "64"],
+        ["--workload", "rf-fit", "--steps", "1", "--warmup", "0",
+         "--rf-fit-rows", "1000", "--rf-fit-trees", "2"],
+    ],
+    ids=["rf", "knn", "svc-fit", "serve", "rf-fit"],
+)
+def test_world8_dress_rehearsal(extra):
+    """Every workload's control flow at world size 8 (gloo, tiny shapes) so
+    the first 8x MI355X driver run has no untested branch (VERDICT r01 #9)."""
+    _run_multirank(extra, 8)

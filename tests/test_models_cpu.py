@@ -285,3 +285,16 @@ def test_predict_proba_parity(split):
     p = lr.predict_proba(Xte)
     np.testing.assert_allclose(p.sum(axis=1), 1.0, atol=1e-9)
     assert (p.argmax(axis=1) == np.searchsorted(np.unique(ytr), lr.predict(Xte))).mean() > 0.999
+
+
+def test_knn_proba_small_reference_sums_to_one():
+    """kneighbors() clamps k to the fit-row count; probabilities must divide
+    by the EFFECTIVE k so rows sum to 1."""
+    from traffic_classifier_sdn_amd.models import KNeighborsClassifier
+
+    rng = np.random.default_rng(3)
+    X = rng.normal(size=(3, 12))  # fewer rows than n_neighbors=5
+    y = np.array(["a", "b", "a"], dtype=object)
+    m = KNeighborsClassifier(n_neighbors=5).fit(X, y)
+    p = m.predict_proba(rng.normal(size=(7, 12)))
+    np.testing.assert_allclose(p.sum(axis=1), 1.0, atol=1e-12)

@@ -80,20 +80,24 @@ class KNeighborsClassifier(Estimator):
         """Uniform-weight neighbour vote fractions (sklearn KNN API)."""
         Xt = as_tensor(X, self.device, torch.float32)
         C = len(self.classes_)
-        k = self.n_neighbors
         dist_k, idx_k = self.kneighbors(Xt)
-        lab = self.y_[idx_k if not self.sharded_ else idx_k.clamp(min=0)]
         if self.sharded_:
             from ..parallel import dist as d
 
             lab_k = self.y_[idx_k]
             cand_d = torch.cat(d.allgather(dist_k), dim=1)
             cand_l = torch.cat(d.allgather(lab_k), dim=1)
+            k = min(self.n_neighbors, cand_d.shape[1])
             _, order = torch.topk(cand_d, k, dim=1, largest=False, sorted=True)
             lab = torch.gather(cand_l, 1, order)
+        else:
+            lab = self.y_[idx_k]
         counts = torch.zeros(Xt.shape[0], C, dtype=torch.float64, device=Xt.device)
         counts.scatter_add_(1, lab.long(), torch.ones_like(lab, dtype=torch.float64))
-        return (counts / k).cpu().numpy()
+        # divide by the EFFECTIVE k: kneighbors() clamps k to the number of
+        # fit rows, so rows must still sum to 1 when the reference set is
+        # smaller than n_neighbors (ADVICE r01)
+        return (counts / lab.shape[1]).cpu().numpy()
 
     # -- checkpointing -------------------------------------------------
     def to_params(self) -> Dict[str, Any]:

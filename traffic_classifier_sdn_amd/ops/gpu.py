@@ -33,23 +33,27 @@ def _f32(t: torch.Tensor) -> torch.Tensor:
 # Model parameters are fixed between fits, but a `.to(float32)` per predict
 # call launches a conversion kernel every time — inside the hipGraph-captured
 # serve path that was ~9 stray elementwise kernels per replay.  Cache the
-# converted copies keyed by source buffer identity (a refit allocates new
-# tensors, so stale hits are not possible without in-place mutation, which
-# the models never do).
-_cast_cache: Dict[Tuple[int, int, torch.dtype], torch.Tensor] = {}
+# converted copies keyed by source buffer address, and keep a reference to
+# the SOURCE tensor in the entry: torch's caching allocator reuses freed
+# addresses, so a data_ptr key alone could serve a stale cast after a model
+# is freed and a same-shape tensor lands at the recycled address (ADVICE
+# r01).  Pinning the source keeps its address out of the free pool for the
+# lifetime of the entry, which makes the (ptr, numel, dtype) key unambiguous
+# (the models never mutate parameters in place).
+_cast_cache: Dict[Tuple[int, int, torch.dtype], Tuple[torch.Tensor, torch.Tensor]] = {}
 
 
 def _f32_cached(t: torch.Tensor) -> torch.Tensor:
     if t.dtype == torch.float32 and t.is_contiguous():
         return t
     key = (t.data_ptr(), t.numel(), t.dtype)
-    c = _cast_cache.get(key)
-    if c is None:
+    ent = _cast_cache.get(key)
+    if ent is None:
         if len(_cast_cache) > 64:
             _cast_cache.clear()
-        c = t.to(torch.float32).contiguous()
-        _cast_cache[key] = c
-    return c
+        ent = (t, t.to(torch.float32).contiguous())
+        _cast_cache[key] = ent
+    return ent[1]
 
 
 def _f64(t: torch.Tensor) -> torch.Tensor:
@@ -67,7 +71,10 @@ def linear_argmax(X: torch.Tensor, coef: torch.Tensor, intercept: torch.Tensor) 
     return _ext.linear_argmax(_f32(X), _f32_cached(coef), _f32_cached(intercept))
 
 
-_gnb_cache: Dict[Tuple[int, int], Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = {}
+# entry = (var_ref, prior_ref, theta32, inv_var, const32): the first two pin
+# the source buffers so their addresses cannot be recycled while cached
+# (same aliasing hazard as _cast_cache)
+_gnb_cache: Dict[Tuple[int, int], Tuple[torch.Tensor, ...]] = {}
 
 
 def gnb_argmax(
@@ -84,9 +91,9 @@ def gnb_argmax(
             _gnb_cache.clear()
         var64 = var.double()
         const = (torch.log(class_prior.double()) - 0.5 * torch.log(2.0 * torch.pi * var64).sum(dim=1))
-        ent = (_f32(theta), (1.0 / var64).float().contiguous(), const.float().contiguous())
+        ent = (var, class_prior, _f32(theta), (1.0 / var64).float().contiguous(), const.float().contiguous())
         _gnb_cache[key] = ent
-    theta32, inv_var, const32 = ent
+    _, _, theta32, inv_var, const32 = ent
     return _ext.gnb_predict(_f32(X), theta32, inv_var, const32)
 
 
@@ -104,20 +111,22 @@ def kmeans_assign(
 # launch simplicity wins; above it the v_mfma_f32_32x32x2_f32 distance
 # GEMM dominates (csrc/knn_mfma.hip)
 _KNN_MFMA_MIN_ROWS = 100_000
-_knn_cmean_cache: Dict[Tuple[int, int], torch.Tensor] = {}
+# entry = (R_ref, cmean): R_ref pins the reference buffer against address
+# recycling
+_knn_cmean_cache: Dict[Tuple[int, int], Tuple[torch.Tensor, torch.Tensor]] = {}
 
 
 def _knn_cmean(R: torch.Tensor) -> torch.Tensor:
     """Reference column means (centering makes the expanded-form distance
     cancellation-safe); cached per reference buffer."""
     key = (R.data_ptr(), R.shape[0])
-    c = _knn_cmean_cache.get(key)
-    if c is None:
+    ent = _knn_cmean_cache.get(key)
+    if ent is None:
         if len(_knn_cmean_cache) > 16:
             _knn_cmean_cache.clear()
-        c = R.mean(dim=0).float().contiguous()
-        _knn_cmean_cache[key] = c
-    return c
+        ent = (R, R.mean(dim=0).float().contiguous())
+        _knn_cmean_cache[key] = ent
+    return ent[1]
 
 
 def _knn_shards(nr: int) -> int:
@@ -178,6 +187,8 @@ def svc_predict(
 
 
 def rf_hist(bins: torch.Tensor, y: torch.Tensor, nid: torch.Tensor, n_nodes: int, n_classes: int) -> torch.Tensor:
+    if bins.shape[1] != 12:  # HIP kernel is specialised for the 12-feature schema
+        return _cpu.rf_hist(bins, y, nid, n_nodes, n_classes)
     hist = torch.zeros(n_nodes, 12, 256, n_classes, dtype=torch.int32, device=bins.device)
     _ext.rf_hist(bins.contiguous(), y.contiguous(), nid.contiguous(), hist)
     return hist

@@ -104,10 +104,24 @@ def _emit(rank, metric, value, unit, world, args, ms_per_step, scaling, config):
 
 def bench_rf(args, rank, world, device, use_gpu):
     model = load_model(RF_CKPT, device=device)
-    # measured (non-timed) accuracy on the real data, reference split
+    # measured (non-timed) accuracies, reference split protocol.  The
+    # reference's 6_quake_training_data.csv is NOT shipped (SURVEY §2.1
+    # C11), so the shipped rows cover 5 classes; the 6-class number refits
+    # on shipped rows + synthetic quake (D-ITG Quake3 replay through the
+    # real collection path) and is labelled _quake_synth accordingly.
     X_real, y_real = load_reference_dataset()
     _, Xte, _, yte = train_test_split_ref(X_real, y_real)
-    acc = accuracy(yte, model.predict(Xte))
+    acc5 = accuracy(yte, model.predict(Xte))
+    from traffic_classifier_sdn_amd.models import RandomForestClassifier
+    from traffic_classifier_sdn_amd.utils.datasets import load_six_class_dataset
+
+    X6, y6 = load_six_class_dataset(seed=args.seed)
+    Xtr6, Xte6, ytr6, yte6 = train_test_split_ref(X6, y6)
+    m6 = RandomForestClassifier(
+        n_estimators=100, seed=args.seed,
+        builder="hist" if use_gpu else "exact", device=device,
+    ).fit(Xtr6, ytr6)
+    acc6 = accuracy(yte6, m6.predict(Xte6))
 
     Xn = synthetic_flow_rows(args.rows_per_gpu, seed=args.seed + 1000 * rank, reference_X=X_real)
     X = torch.from_numpy(Xn).to(device)
@@ -130,8 +144,10 @@ def bench_rf(args, rank, world, device, use_gpu):
             "global_batch": args.rows_per_gpu * world,
             "seq_len": 12,
             "parallelism": f"dp{world}",
-            "accuracy_6class": acc,
-            "accuracy_published_ref": 0.9987,
+            "accuracy_6class_quake_synth": acc6,
+            "accuracy_5class_shipped": acc5,
+            "accuracy_published_ref_6class": 0.9987,
+            "quake_note": "reference's 6_quake_training_data.csv (1244 rows) is not shipped; 6th class synthesized via D-ITG Quake3 replay (utils.datasets.synthesize_quake_rows)",
             "dtype_note": "f32 features/thresholds, prediction-parity-tested vs f64 sklearn oracles; fit paths accumulate f64",
         },
     )

@@ -19,6 +19,8 @@ CASES = {
     "rf": ["--steps", "2", "--warmup", "1"],
     "knn": ["--workload", "knn", "--steps", "2", "--warmup", "1", "--knn-queries", "2048"],
     "svc-fit": ["--workload", "svc-fit", "--steps", "2", "--warmup", "1", "--svc-iters-per-step", "5"],
+    "svc-fit-full": ["--workload", "svc-fit-full", "--steps", "1", "--warmup", "0",
+                     "--svc-rows", "1500", "--svc-full-max-iter", "60"],
     "serve": ["--workload", "serve", "--steps", "3", "--warmup", "1"],
     "rf-fit": ["--workload", "rf-fit", "--steps", "2", "--warmup", "1"],
 }
@@ -78,11 +80,13 @@ def test_gpus_flag_spawns_ranks():
          "--knn-queries", "256", "--knn-ref-rows-per-gpu", "1000"],
         ["--workload", "svc-fit", "--steps", "2", "--warmup", "1",
          "--svc-iters-per-step", "3", "--svc-rows", "4000"],
+        ["--workload", "svc-fit-full", "--steps", "1", "--warmup", "0",
+         "--svc-rows", "2400", "--svc-full-max-iter", "30"],
         ["--workload", "serve", "--steps", "2", "--warmup", "1", "--serve-flows", "64"],
         ["--workload", "rf-fit", "--steps", "1", "--warmup", "0",
          "--rf-fit-rows", "1000", "--rf-fit-trees", "2"],
     ],
-    ids=["rf", "knn", "svc-fit", "serve", "rf-fit"],
+    ids=["rf", "knn", "svc-fit", "svc-fit-full", "serve", "rf-fit"],
 )
 def test_world8_dress_rehearsal(extra):
     """Every workload's control flow at world size 8 (gloo, tiny shapes) so

@@ -62,6 +62,14 @@ extern "C" void launch_gnb_fit_stats(const double* X, const long long* y,
 // Per-row softmax in registers (C <= 16), per-block LDS gradient [C, F+1]
 // f64, one global atomic sweep per block.  The l2 term is applied on host.
 // ---------------------------------------------------------------------------
+// Accumulation strategy (round-2 rewrite): the first version did C*(F+1)
+// (=78 at C=6) LDS f64 atomicAdds PER ROW, and within a wave all 64 lanes
+// hit the SAME 78 addresses — a 64-way serialized conflict chain that
+// measured 65% VALUBusy / ~1 ms per 1M-row pass (profiles/
+// pmc_counters_r02.md).  Now each entry of the (p - onehot) ⊗ [x,1] outer
+// product is wave-reduced in registers (__shfl_down tree, guide G12:
+// per-wave partial reduction first) and lane 0 adds it to a PER-WAVE LDS
+// slice — zero atomics in the hot loop, one global atomic sweep per block.
 template <int C>
 __global__ void logistic_grad_kernel(const double* __restrict__ X,
                                      const long long* __restrict__ y,
@@ -71,60 +79,83 @@ __global__ void logistic_grad_kernel(const double* __restrict__ X,
                                      double* __restrict__ loss,  // [1]
                                      long long n) {
   constexpr int F = 12;
+  constexpr int NG = C * (F + 1);
   extern __shared__ __attribute__((aligned(16))) char smem[];
   double* s_w = reinterpret_cast<double*>(smem);  // [C,F]
   double* s_b = s_w + C * F;                      // [C]
-  double* s_g = s_b + C;                          // [C,F+1]
+  double* s_g = s_b + C;                          // [nwaves][NG]
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int nwaves = blockDim.x / WAVE;
   for (int i = threadIdx.x; i < C * F; i += blockDim.x) s_w[i] = W[i];
   for (int i = threadIdx.x; i < C; i += blockDim.x) s_b[i] = b[i];
-  for (int i = threadIdx.x; i < C * (F + 1); i += blockDim.x) s_g[i] = 0.0;
+  for (int i = threadIdx.x; i < nwaves * NG; i += blockDim.x) s_g[i] = 0.0;
   __syncthreads();
 
   double local_loss = 0.0;
+  double* gw = s_g + wid * NG;
   long long stride = (long long)gridDim.x * blockDim.x;
-  for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x; row < n;
-       row += stride) {
+  // uniform trip count per block: every lane reaches the wave_sum calls
+  for (long long base = (long long)blockIdx.x * blockDim.x; base < n;
+       base += stride) {
+    long long row = base + threadIdx.x;
+    bool valid = row < n;
     double xv[F];
+    double pv[C];
 #pragma unroll
-    for (int j = 0; j < F; ++j) xv[j] = X[row * F + j];
-    double logits[C];
-    double m = -INFINITY;
+    for (int j = 0; j < F; ++j) xv[j] = 0.0;
 #pragma unroll
-    for (int c = 0; c < C; ++c) {
-      double s = s_b[c];
+    for (int c = 0; c < C; ++c) pv[c] = 0.0;
+    if (valid) {
 #pragma unroll
-      for (int j = 0; j < F; ++j) s += xv[j] * s_w[c * F + j];
-      logits[c] = s;
-      m = fmax(m, s);
+      for (int j = 0; j < F; ++j) xv[j] = X[row * F + j];
+      double logits[C];
+      double m = -INFINITY;
+#pragma unroll
+      for (int c = 0; c < C; ++c) {
+        double s = s_b[c];
+#pragma unroll
+        for (int j = 0; j < F; ++j) s += xv[j] * s_w[c * F + j];
+        logits[c] = s;
+        m = fmax(m, s);
+      }
+      int yc = (int)y[row];
+      double z = 0.0;
+      // keep logit_y - m BEFORE the exp: log(exp(t)) underflows to -inf for
+      // t < -745, which raw-scale flow features reach easily once the
+      // weights grow — the CPU oracle's log_softmax never exponentiates the
+      // margin
+      double ly = 0.0;
+#pragma unroll
+      for (int c = 0; c < C; ++c) {
+        double t = logits[c] - m;
+        if (c == yc) ly = t;
+        logits[c] = exp(t);
+        z += logits[c];
+      }
+      local_loss += -(ly - log(z));
+      double inv_z = 1.0 / z;
+#pragma unroll
+      for (int c = 0; c < C; ++c)
+        pv[c] = logits[c] * inv_z - (c == yc ? 1.0 : 0.0);
     }
-    int yc = (int)y[row];
-    double z = 0.0;
-    // keep logit_y - m BEFORE the exp: log(exp(t)) underflows to -inf for
-    // t < -745, which raw-scale flow features reach easily once the weights
-    // grow — the CPU oracle's log_softmax never exponentiates the margin
-    double ly = 0.0;
 #pragma unroll
     for (int c = 0; c < C; ++c) {
-      double t = logits[c] - m;
-      if (c == yc) ly = t;
-      logits[c] = exp(t);
-      z += logits[c];
-    }
-    local_loss += -(ly - log(z));
-    double inv_z = 1.0 / z;
 #pragma unroll
-    for (int c = 0; c < C; ++c) {
-      double p = logits[c] * inv_z - (c == yc ? 1.0 : 0.0);
-#pragma unroll
-      for (int j = 0; j < F; ++j) atomicAdd(&s_g[c * (F + 1) + j], p * xv[j]);
-      atomicAdd(&s_g[c * (F + 1) + F], p);
+      for (int j = 0; j <= F; ++j) {
+        double v = wave_sum(j < F ? pv[c] * xv[j] : pv[c]);
+        if (lane == 0) gw[c * (F + 1) + j] += v;
+      }
     }
   }
   local_loss = wave_sum(local_loss);
-  if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(loss, local_loss);
+  if (lane == 0) atomicAdd(loss, local_loss);
   __syncthreads();
-  for (int i = threadIdx.x; i < C * (F + 1); i += blockDim.x)
-    atomicAdd(&grad[i], s_g[i]);
+  for (int i = threadIdx.x; i < NG; i += blockDim.x) {
+    double s = 0.0;
+    for (int w = 0; w < nwaves; ++w) s += s_g[w * NG + i];
+    atomicAdd(&grad[i], s);
+  }
 }
 
 extern "C" void launch_logistic_grad(const double* X, const long long* y,
@@ -132,7 +163,7 @@ extern "C" void launch_logistic_grad(const double* X, const long long* y,
                                      double* grad, double* loss, long long n,
                                      int C, hipStream_t stream) {
   const int block = 256;
-  size_t bytes = (size_t)(C * 12 + C + C * 13) * sizeof(double);
+  size_t bytes = (size_t)(C * 12 + C + (block / 64) * C * 13) * sizeof(double);
   dim3 grid(ts_grid(n, block));
 #define LG_CASE(CV)                                                         \
   case CV:                                                                  \

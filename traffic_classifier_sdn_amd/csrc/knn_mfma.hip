@@ -9,24 +9,33 @@
 // around the reference column means c, which removes the catastrophic
 // cancellation raw byte/packet-count features would cause (translation
 // leaves Euclidean distance invariant; the per-query constant drops out of
-// the ordering).  Selection is a per-row threshold filter: producing lanes
-// compare each key against a broadcast (stale-tolerant, conservative)
-// per-row threshold and only SURVIVORS spill into a per-wave LDS key
-// matrix + survivor bitmask; the owning lane pair then merges them into a
-// register-resident exact top-k list.  Stale thresholds only admit extra
-// survivors (the owner re-checks), never drop one, so the selection is
-// exact; in steady state a candidate costs one compare beyond the MFMA.
-// The final k winners per query are REFINED with the exact
-// direct-difference f32 distance (same numerics as the scalar kernel).
+// the ordering).
+//
+// Selection (round-2 SWAPPED-OPERAND design): the MFMA computes
+// D[cand][query] (A = centered R-tile, B = centered Q^T), so the query
+// index lands in the LANE (col = lane&31) and each lane's 16 accumulator
+// registers are 16 CANDIDATES of its own query — top-k selection is pure
+// lane-local register work (one fmaf + compare per candidate, rare sorted
+// insert).  Round 1 computed D[query][cand] (A = Q), which scattered one
+// query's candidates across the wave and needed a per-wave LDS key matrix,
+// survivor bitmask, owner scan, and stale-threshold publishes — measured
+// epilogue-bound at 42.4% MfmaUtil (profiles/pmc_counters_r01.md).  The
+// swap deletes all of that LDS machinery (-18.4 KB/WG, no atomics, no wave
+// barriers in the hot loop).  The final k winners per query are REFINED
+// with the exact direct-difference f32 distance (same numerics as the
+// scalar kernel).
 //
 // Grid: (ceil(nq/QB), S) — query blocks x reference shards; knn_merge_kernel
 // folds the S partial lists per query (+ fused uniform vote).
 //
 // Per workgroup (256 threads = 4 waves):
 //   QB=256 queries staged centered+transposed in LDS (12 KB)
-//   R streamed in 128-candidate tiles, centered+transposed (6 KB) + norms
-//   wave w owns query tiles w, w+4 (32 rows each; row r of a tile is
-//   owned by lanes r and r+32, scanning key columns 0-15 / 16-31)
+//   R streamed in 256-candidate tiles, double-buffered centered+transposed
+//   (2 x 12 KB) + norms — ONE __syncthreads per tile (stage t+1 into the
+//   idle buffer while computing t)
+//   wave w serves query groups w, w+4 (32 queries each; lane pair
+//   (l, l+32) shares query (group*32 + l&31) and splits each 32-candidate
+//   subtile between its two accumulator row maps)
 
 #include <hip/hip_runtime.h>
 
@@ -35,10 +44,9 @@
 #include "common.h"
 
 #define KM_QB 256      // queries per workgroup (8 query tiles, 2 per wave)
-#define KM_TB 128      // candidate tile (4 MFMA column-subtiles)
+#define KM_TB 256      // candidate tile (8 MFMA column-subtiles)
 #define KM_KMAX 8      // max k supported by this path
 #define KM_F 12
-#define KM_PITCH 33    // keymat row pitch (bank-staggered)
 
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 
@@ -69,21 +77,13 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
     int* __restrict__ part_i,         // [S, nq, k] global candidate index (-1 pad)
     long long nq, long long nr, int k, long long shard_rows) {
   __shared__ float s_qt[KM_F][KM_QB];            // centered Q^T
-  __shared__ float s_rt[KM_F][KM_TB];            // centered R-tile^T
-  __shared__ float s_rn[KM_TB];                  // ||r-c||^2 (FLT_MAX pad)
-  // per-wave survivor key matrix [col][row] + per-row survivor bitmask:
-  // producers only spill keys that beat the (stale-tolerant) per-row
-  // threshold, so in steady state a subtile costs 16 broadcast threshold
-  // reads + compares and the owner scan reads nothing at all — the LDS
-  // pipe stops being the bottleneck (was: full 16-write/16-read key spill)
-  __shared__ float s_km[4][32 * KM_PITCH];
-  __shared__ unsigned s_mask[4][32];
-  __shared__ float s_worst[KM_QB];  // per-row k-th-best key (owners update)
+  __shared__ float s_rt[2][KM_F][KM_TB];         // centered R-tile^T (2 bufs)
+  __shared__ float s_rn[2][KM_TB];               // ||r-c||^2 (FLT_MAX pad)
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wave = tid >> 6;
-  const int half = lane >> 5;  // MFMA k-index; also this lane's column half
+  const int half = lane >> 5;  // MFMA k-index (K=2 per instruction)
   const int l31 = lane & 31;
   const long long qb0 = (long long)blockIdx.x * KM_QB;
   const int shard = blockIdx.y;
@@ -111,15 +111,31 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
     Row12 x = load_row12(Q, q);
 #pragma unroll
     for (int j = 0; j < KM_F; ++j) s_qt[j][i] = x.v[j] - cm[j];
-    s_worst[i] = FLT_MAX;
   }
-  if (tid < 4)
-#pragma unroll
-    for (int r = 0; r < 32; ++r) s_mask[tid][r] = 0u;
   __syncthreads();
 
-  // per-(lane, query-tile) top-k sub-list in registers; the lane pair
-  // (r, r+32) covers key columns [0,16) / [16,32) of row r
+  // SWAPPED-OPERAND selection (round 2): with A = R-tile and B = Q^T the
+  // D[cand][query] mapping puts QUERY in the lane index (col = lane&31) and
+  // 16 CANDIDATES in each lane's accumulator registers — so top-k selection
+  // is pure lane-local register work (compare + rare sorted insert).  The
+  // round-1 layout (A = Q) spread one query's candidates across lanes and
+  // needed an LDS key matrix + survivor bitmask + owner scan + stale
+  // threshold publishes (42.4% MfmaUtil, epilogue-bound).  All of that is
+  // gone: no LDS spill, no atomics, no wave barriers in the hot loop.
+  //
+  // Lane l serves query qb0 + (wave + 4*qti)*32 + (l&31); the lane pair
+  // (l, l+32) holds the same query and splits each 32-candidate subtile by
+  // the D-row map (rowmap(g, l>>5)).  The merge/refine tail below is the
+  // same lane-pair contract as round 1.
+  float bfrag[2][6];  // Q fragments per query group, loaded once
+#pragma unroll
+  for (int qti = 0; qti < 2; ++qti) {
+    const int qbase = (wave + 4 * qti) * 32;
+#pragma unroll
+    for (int s = 0; s < 6; ++s) bfrag[qti][s] = s_qt[2 * s + half][qbase + l31];
+  }
+
+  // per-(lane, query-group) exact top-k sub-list in registers
   float lk[2][KM_KMAX];
   int li[2][KM_KMAX];
   float wkey[2];
@@ -135,119 +151,84 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
 
   // ---- main loop over candidate tiles ------------------------------------
   // Software prefetch: each tile's global row load is issued one iteration
-  // ahead (into registers), so its ~HBM round-trip overlaps the previous
-  // tile's MFMA work instead of stalling the whole workgroup at staging.
-  const int pf_i = tid;  // thread -> candidate slot (KM_TB <= blockDim.x)
+  // ahead (into registers); LDS staging is double-buffered so ONE
+  // __syncthreads per tile separates write(buf) from compute(buf) — the
+  // barrier of the NEXT iteration separates compute(buf) from the write
+  // that reuses buf two tiles later.
+  const int pf_i = tid;  // thread -> candidate slot (KM_TB == blockDim.x)
   Row12 pf_row;
   bool pf_valid = false;
-  if (pf_i < KM_TB && r0 + pf_i < r1) {
+  if (r0 + pf_i < r1) {
     pf_row = load_row12(R, r0 + pf_i);
     pf_valid = true;
   }
-  for (long long tb = r0; tb < r1; tb += KM_TB) {
-    const int cnt_t = (int)min((long long)KM_TB, r1 - tb);
-    __syncthreads();  // previous tile fully consumed
-    if (pf_i < KM_TB) {
-      if (pf_valid) {
-        float rn = 0.f;
+  int t = 0;
+  for (long long tb = r0; tb < r1; tb += KM_TB, ++t) {
+    const int buf = t & 1;
+    // stage tile t (prefetched last iteration) into its buffer
+    if (pf_valid) {
+      float rn = 0.f;
 #pragma unroll
-        for (int j = 0; j < KM_F; ++j) {
-          float rc = pf_row.v[j] - cm[j];
-          s_rt[j][pf_i] = rc;
-          rn = fmaf(rc, rc, rn);
-        }
-        s_rn[pf_i] = rn;
-      } else {
-#pragma unroll
-        for (int j = 0; j < KM_F; ++j) s_rt[j][pf_i] = 0.f;
-        s_rn[pf_i] = FLT_MAX;  // padded candidate never selected
+      for (int j = 0; j < KM_F; ++j) {
+        float rc = pf_row.v[j] - cm[j];
+        s_rt[buf][j][pf_i] = rc;
+        rn = fmaf(rc, rc, rn);
       }
+      s_rn[buf][pf_i] = rn;
+    } else {
+#pragma unroll
+      for (int j = 0; j < KM_F; ++j) s_rt[buf][j][pf_i] = 0.f;
+      s_rn[buf][pf_i] = FLT_MAX;  // padded candidate never selected
     }
-    __syncthreads();
-    // issue the NEXT tile's loads now; the waitcnt lands at the next
+    // issue the NEXT tile's loads now; their waitcnt lands at the next
     // iteration's staging writes, hidden behind this tile's compute
     {
       long long nxt = tb + KM_TB + pf_i;
-      pf_valid = (pf_i < KM_TB) && (nxt < r1);
+      pf_valid = nxt < r1;
       if (pf_valid) pf_row = load_row12(R, nxt);
     }
+    __syncthreads();  // tile t visible; buf^1 free for the next staging pass
 
-#pragma unroll
-    for (int qti = 0; qti < 2; ++qti) {
-      const int qt = wave + 4 * qti;
-      const int rowbase = qt * 32;
-      // A fragments: lane l -> Qc[rowbase + (l&31)][2s + (l>>5)]
+    const float(*rt)[KM_TB] = s_rt[buf];
+    const float* rn_t = s_rn[buf];
+
+    for (int ct = 0; ct < KM_TB / 32; ++ct) {
+      // A fragments: candidate rows, lane l -> Rc[ct*32 + (l&31)][2s + half];
+      // loaded ONCE per subtile and reused by both query groups
       float afrag[6];
 #pragma unroll
-      for (int s = 0; s < 6; ++s) afrag[s] = s_qt[2 * s + half][rowbase + l31];
-      // per-row threshold cache (broadcast reads); stale values only admit
-      // extra survivors — the owner re-checks against its exact register
-      // worst — never miss one
-      float tau[16];
+      for (int s = 0; s < 6; ++s) afrag[s] = rt[2 * s + half][ct * 32 + l31];
 #pragma unroll
-      for (int g = 0; g < 16; ++g) tau[g] = s_worst[rowbase + km_rowmap(g, half)];
-
-      for (int ct = 0; ct < KM_TB / 32; ++ct) {
+      for (int qti = 0; qti < 2; ++qti) {
         f32x16 acc = {};
 #pragma unroll
         for (int s = 0; s < 6; ++s)
-          acc = __builtin_amdgcn_mfma_f32_32x32x2f32(
-              afrag[s], s_rt[2 * s + half][ct * 32 + l31], acc, 0, 0, 0);
-        const float rncol = s_rn[ct * 32 + l31];
-        // survivors spill (key -> own slot, col bit -> row mask); padded
-        // columns carry key = FLT_MAX and never pass
+          acc = __builtin_amdgcn_mfma_f32_32x32x2f32(afrag[s], bfrag[qti][s],
+                                                     acc, 0, 0, 0);
+        // lane-local selection: 16 candidates of THIS lane's query; padded
+        // candidates carry rn = FLT_MAX and never pass the compare
 #pragma unroll
         for (int g = 0; g < 16; ++g) {
-          float key = fmaf(-2.f, acc[g], rncol);
-          if (key < tau[g]) {
-            int row = km_rowmap(g, half);
-            s_km[wave][l31 * KM_PITCH + row] = key;
-            atomicOr(&s_mask[wave][row], 1u << l31);
+          const int cand = ct * 32 + km_rowmap(g, half);
+          float key = fmaf(-2.f, acc[g], rn_t[cand]);
+          if (key < wkey[qti]) {
+            int ws = 0;
+            float wv = -FLT_MAX;
+#pragma unroll
+            for (int j = 0; j < KM_KMAX; ++j)
+              if (j < k && lk[qti][j] > wv) {
+                wv = lk[qti][j];
+                ws = j;
+              }
+            lk[qti][ws] = key;
+            li[qti][ws] = (int)(tb + cand);
+            wv = -FLT_MAX;
+#pragma unroll
+            for (int j = 0; j < KM_KMAX; ++j)
+              if (j < k && lk[qti][j] > wv) wv = lk[qti][j];
+            wkey[qti] = wv;
           }
         }
-        __builtin_amdgcn_wave_barrier();
-        // owner scan: lane pair (l31, l31+32) splits the survivor bits of
-        // row l31 (low/high 16 columns)
-        {
-          const int row = l31;
-          unsigned m = s_mask[wave][row];
-          unsigned mh = half ? (m >> 16) : (m & 0xffffu);
-          const long long colbase = tb + ct * 32 + half * 16;
-          while (mh) {
-            int cc = __ffs(mh) - 1;
-            mh &= mh - 1;
-            float key = s_km[wave][(half * 16 + cc) * KM_PITCH + row];
-            if (key < wkey[qti]) {
-              int ws = 0;
-              float wv = -FLT_MAX;
-#pragma unroll
-              for (int j = 0; j < KM_KMAX; ++j)
-                if (j < k && lk[qti][j] > wv) {
-                  wv = lk[qti][j];
-                  ws = j;
-                }
-              lk[qti][ws] = key;
-              li[qti][ws] = (int)(colbase + cc);
-              wv = -FLT_MAX;
-#pragma unroll
-              for (int j = 0; j < KM_KMAX; ++j)
-                if (j < k && lk[qti][j] > wv) wv = lk[qti][j];
-              wkey[qti] = wv;
-            }
-          }
-          // a produced column can land in either half's sub-list, so the
-          // shared threshold must be conservative for BOTH: publish the max
-          // of the lane pair's worsts (stale/loose admits extra survivors,
-          // never drops one)
-          float other = __shfl(wkey[qti], lane ^ 32, WAVE);
-          float pub = fmaxf(wkey[qti], other);
-          __builtin_amdgcn_wave_barrier();
-          if (half == 0) {
-            s_worst[rowbase + row] = pub;
-            if (m) s_mask[wave][row] = 0u;
-          }
-        }
-        __builtin_amdgcn_wave_barrier();
       }
     }
   }

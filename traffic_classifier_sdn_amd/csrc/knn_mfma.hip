@@ -198,6 +198,16 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
       float afrag[6];
 #pragma unroll
       for (int s = 0; s < 6; ++s) afrag[s] = rt[2 * s + half][ct * 32 + l31];
+      // hoist this lane's 16 candidate norms as 4 batched ds_read_b128
+      // (rowmap hits 4 runs of 4 consecutive rows: {0,8,16,24}+4*half);
+      // leaving the reads inline made hipcc serialize ~50-cycle LDS
+      // latency per candidate check — 32 dependent waits per subtile
+      const float4* rn4 =
+          reinterpret_cast<const float4*>(rn_t + ct * 32 + 4 * half);
+      float4 rnv[4];
+#pragma unroll
+      for (int r4 = 0; r4 < 4; ++r4) rnv[r4] = rn4[2 * r4];
+      const float* rnf = reinterpret_cast<const float*>(rnv);
 #pragma unroll
       for (int qti = 0; qti < 2; ++qti) {
         f32x16 acc = {};
@@ -206,27 +216,39 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
           acc = __builtin_amdgcn_mfma_f32_32x32x2f32(afrag[s], bfrag[qti][s],
                                                      acc, 0, 0, 0);
         // lane-local selection: 16 candidates of THIS lane's query; padded
-        // candidates carry rn = FLT_MAX and never pass the compare
+        // candidates carry rn = FLT_MAX and never pass the compare.  One
+        // min-tree + ONE guarded branch per query group instead of 16
+        // per-candidate skip branches: the common (no-survivor) path is 16
+        // fma + 15 min + 1 compare, all VALU, and the insert block's exec
+        // mask is wave-collective (taken only when SOME lane has a winner).
+        float keyv[16];
 #pragma unroll
-        for (int g = 0; g < 16; ++g) {
-          const int cand = ct * 32 + km_rowmap(g, half);
-          float key = fmaf(-2.f, acc[g], rn_t[cand]);
-          if (key < wkey[qti]) {
-            int ws = 0;
-            float wv = -FLT_MAX;
+        for (int g = 0; g < 16; ++g) keyv[g] = fmaf(-2.f, acc[g], rnf[g]);
+        float kmin = keyv[0];
 #pragma unroll
-            for (int j = 0; j < KM_KMAX; ++j)
-              if (j < k && lk[qti][j] > wv) {
-                wv = lk[qti][j];
-                ws = j;
-              }
-            lk[qti][ws] = key;
-            li[qti][ws] = (int)(tb + cand);
-            wv = -FLT_MAX;
+        for (int g = 1; g < 16; ++g) kmin = fminf(kmin, keyv[g]);
+        if (kmin < wkey[qti]) {
 #pragma unroll
-            for (int j = 0; j < KM_KMAX; ++j)
-              if (j < k && lk[qti][j] > wv) wv = lk[qti][j];
-            wkey[qti] = wv;
+          for (int g = 0; g < 16; ++g) {
+            float key = keyv[g];
+            if (key < wkey[qti]) {
+              const int cand = ct * 32 + km_rowmap(g, half);
+              int ws = 0;
+              float wv = -FLT_MAX;
+#pragma unroll
+              for (int j = 0; j < KM_KMAX; ++j)
+                if (j < k && lk[qti][j] > wv) {
+                  wv = lk[qti][j];
+                  ws = j;
+                }
+              lk[qti][ws] = key;
+              li[qti][ws] = (int)(tb + cand);
+              wv = -FLT_MAX;
+#pragma unroll
+              for (int j = 0; j < KM_KMAX; ++j)
+                if (j < k && lk[qti][j] > wv) wv = lk[qti][j];
+              wkey[qti] = wv;
+            }
           }
         }
       }

@@ -43,7 +43,8 @@
 
 #include "common.h"
 
-#define KM_QB 256      // queries per workgroup (8 query tiles, 2 per wave)
+#define KM_NQT 2       // query groups per wave
+#define KM_QB (32 * 4 * KM_NQT)  // queries per workgroup
 #define KM_TB 256      // candidate tile (8 MFMA column-subtiles)
 #define KM_KMAX 8      // max k supported by this path
 #define KM_F 12
@@ -127,20 +128,20 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
   // (l, l+32) holds the same query and splits each 32-candidate subtile by
   // the D-row map (rowmap(g, l>>5)).  The merge/refine tail below is the
   // same lane-pair contract as round 1.
-  float bfrag[2][6];  // Q fragments per query group, loaded once
+  float bfrag[KM_NQT][6];  // Q fragments per query group, loaded once
 #pragma unroll
-  for (int qti = 0; qti < 2; ++qti) {
+  for (int qti = 0; qti < KM_NQT; ++qti) {
     const int qbase = (wave + 4 * qti) * 32;
 #pragma unroll
     for (int s = 0; s < 6; ++s) bfrag[qti][s] = s_qt[2 * s + half][qbase + l31];
   }
 
   // per-(lane, query-group) exact top-k sub-list in registers
-  float lk[2][KM_KMAX];
-  int li[2][KM_KMAX];
-  float wkey[2];
+  float lk[KM_NQT][KM_KMAX];
+  int li[KM_NQT][KM_KMAX];
+  float wkey[KM_NQT];
 #pragma unroll
-  for (int t = 0; t < 2; ++t) {
+  for (int t = 0; t < KM_NQT; ++t) {
 #pragma unroll
     for (int j = 0; j < KM_KMAX; ++j) {
       lk[t][j] = FLT_MAX;
@@ -209,7 +210,7 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
       for (int r4 = 0; r4 < 4; ++r4) rnv[r4] = rn4[2 * r4];
       const float* rnf = reinterpret_cast<const float*>(rnv);
 #pragma unroll
-      for (int qti = 0; qti < 2; ++qti) {
+      for (int qti = 0; qti < KM_NQT; ++qti) {
         f32x16 acc = {};
 #pragma unroll
         for (int s = 0; s < 6; ++s)
@@ -258,7 +259,7 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
 
   // ---- merge lane pairs, refine with exact distances, emit sorted --------
 #pragma unroll
-  for (int qti = 0; qti < 2; ++qti) {
+  for (int qti = 0; qti < KM_NQT; ++qti) {
     const int qt = wave + 4 * qti;
     const long long q = qb0 + qt * 32 + l31;
     // pull the partner half's sub-list (lane r+32 -> lane r and vice versa;

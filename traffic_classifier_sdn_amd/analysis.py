@@ -15,8 +15,11 @@ standardize + projection passes over the rows):
 - KMeans on the standardized rows with the mode-based cluster→class
   assignment and its supervised accuracy (notebook cells 104-125)
 
-No plotting: the numbers the notebook reads off its matplotlib figures are
-returned directly (scatter/contour rendering is presentation, not compute).
+``--plots DIR`` additionally writes the notebook's three figures as SVG
+(PCA scatter / LR-on-2PC decision boundary / KMeans cluster scatter —
+notebook cells 85, 98, 126) via the dependency-free utils.svgplot writer;
+without the flag only the numbers the notebook reads off its figures are
+returned.
 """
 
 from __future__ import annotations
@@ -62,7 +65,7 @@ def pca(X: torch.Tensor, n_components: int = 2):
     return comps, ratio[:n_components], proj
 
 
-def run_analysis(device: Optional[str] = None, seed: int = 101) -> Dict:
+def run_analysis(device: Optional[str] = None, seed: int = 101, plots: Optional[str] = None) -> Dict:
     dev = device or ("cuda" if torch.cuda.is_available() else "cpu")
     X, y = load_reference_dataset()
     Xt = torch.as_tensor(X, dtype=torch.float64, device=dev)
@@ -94,6 +97,36 @@ def run_analysis(device: Optional[str] = None, seed: int = 101) -> Dict:
     out["kmeans_n_iter"] = int(km.n_iter_)
     out["kmeans_mode_assignment"] = [str(n) for n in names]
     out["kmeans_supervised_accuracy"] = accuracy(y, names[cluster_ids])
+
+    if plots:
+        import os
+
+        from .utils.svgplot import scatter_svg
+
+        os.makedirs(plots, exist_ok=True)
+        classes = sorted(set(str(v) for v in y))
+        lut = {c: i for i, c in enumerate(classes)}
+        y_idx = np.asarray([lut[str(v)] for v in y])
+        written = [
+            scatter_svg(
+                os.path.join(plots, "pca_scatter.svg"), P, y_idx, classes,
+                "PCA of flow features (true class)",
+            ),
+            scatter_svg(
+                os.path.join(plots, "pca_decision_boundary.svg"), Pte,
+                np.asarray([lut[str(v)] for v in yte]), classes,
+                f"LR on 2 PCs — decision regions (acc {out['lr_accuracy_on_2pc']:.3f})",
+                decision_fn=lambda G: np.asarray(
+                    [np.searchsorted(lr2.classes_.astype(str), str(v)) for v in lr2.predict(G)]
+                ),
+            ),
+            scatter_svg(
+                os.path.join(plots, "kmeans_clusters.svg"), P, cluster_ids,
+                [f"cluster {i} -> {names[i]}" for i in range(km.n_clusters)],
+                "KMeans clusters in PC space (mode-assigned class)",
+            ),
+        ]
+        out["plots"] = written
     return out
 
 
@@ -101,8 +134,10 @@ def main(argv: Optional[Sequence[str]] = None) -> int:
     ap = argparse.ArgumentParser(description=__doc__)
     ap.add_argument("--device", default=None)
     ap.add_argument("--json", action="store_true")
+    ap.add_argument("--plots", default=None, metavar="DIR",
+                    help="write the notebook's three figures as SVG into DIR")
     args = ap.parse_args(argv)
-    res = run_analysis(device=args.device)
+    res = run_analysis(device=args.device, plots=args.plots)
     if args.json:
         print(json.dumps(res))
     else:

@@ -338,3 +338,21 @@ def test_analysis_runs_on_gpu():
     assert res["device"].startswith("cuda")
     assert res["lr_accuracy_on_2pc"] > 0.8
     assert len(res["kmeans_mode_assignment"]) == 6
+
+
+@pytest.mark.gpu
+def test_svc_full_ovo_fit_6class_gpu():
+    """Full 6-class one-vs-one SVC fit (all 15 pairs, device-fused SMO) on
+    stratified synthetic rows — the svc-fit-full bench path (config #3)."""
+    from traffic_classifier_sdn_amd.models import SVC
+    from traffic_classifier_sdn_amd.utils.datasets import load_six_class_dataset
+
+    X6, y6 = load_six_class_dataset(quake_rows=400, seed=2)
+    rng = np.random.default_rng(0)
+    idx = rng.permutation(len(X6))
+    tr, te = idx[:6000], idx[6000:7500]
+    m = SVC(tol=1e-3, max_iter=4000, device="cuda").fit(X6[tr], y6[tr])
+    assert len(m.n_iter_) == 15  # all OVO pairs ran
+    assert int(m.n_support_.sum()) > 0
+    acc = (m.predict(X6[te]).astype(str) == y6[te].astype(str)).mean()
+    assert acc > 0.85, acc  # published 6-class SVC reference accuracy: 85.01

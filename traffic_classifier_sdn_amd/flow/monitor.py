@@ -19,6 +19,7 @@ from __future__ import annotations
 
 import argparse
 import asyncio
+import struct
 import sys
 import time
 from typing import Dict, Optional, TextIO
@@ -105,6 +106,18 @@ class MonitorApp:
             writer.close()
 
     async def _dispatch(self, dp: Datapath, msg_type: int, xid: int, body: bytes) -> None:
+        try:
+            await self._dispatch_inner(dp, msg_type, xid, body)
+        except (ValueError, IndexError, KeyError, struct.error) as e:
+            # a malformed message from one switch must not tear down its
+            # channel (tests/test_openflow_golden.py fuzz coverage); the
+            # framing (header length) already kept the stream in sync
+            self._log_err(f"malformed OF message type={msg_type} xid={xid}: {e!r}")
+
+    def _log_err(self, msg: str) -> None:
+        print(msg, file=sys.stderr)
+
+    async def _dispatch_inner(self, dp: Datapath, msg_type: int, xid: int, body: bytes) -> None:
         if msg_type == of.OFPT_HELLO:
             return
         if msg_type == of.OFPT_ECHO_REQUEST:

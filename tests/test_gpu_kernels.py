@@ -356,3 +356,35 @@ def test_svc_full_ovo_fit_6class_gpu():
     assert int(m.n_support_.sum()) > 0
     acc = (m.predict(X6[te]).astype(str) == y6[te].astype(str)).mean()
     assert acc > 0.85, acc  # published 6-class SVC reference accuracy: 85.01
+
+
+@pytest.mark.gpu
+def test_svc_predict_large_nsv_accumulation():
+    """Regression: the tiled svc_predict kernel accumulated tens of
+    thousands of SIGNED near-unit dual terms in one sequential f32 chain —
+    at 35K SVs (non-separable pair, alphas at the C bound) rounding
+    swallowed the O(1) decision values and flipped 24% of votes at
+    n>32768 rows (the tiled-kernel regime).  Now per-tile f32 partials fold
+    into f64 accumulators.  Oracle: f64 torch decision."""
+    from traffic_classifier_sdn_amd.ops.gpu import _ext
+
+    rng = np.random.default_rng(11)
+    C, nsv_per, n = 6, 6000, 40_000  # n > 32768 -> tiled kernel
+    nsv = C * nsv_per
+    SV = torch.from_numpy(rng.normal(size=(nsv, 12))).float()
+    # bounded-alpha-style duals: mostly +-1 (the pathological regime)
+    dual = torch.from_numpy(
+        rng.choice([-1.0, 1.0], size=(C - 1, nsv)) * rng.uniform(0.5, 1.0, size=(C - 1, nsv))
+    ).float()
+    n_support = torch.full((C,), nsv_per, dtype=torch.int64)
+    svclass = torch.repeat_interleave(torch.arange(C), nsv_per).to(torch.uint8)
+    intercept = torch.from_numpy(rng.normal(size=15)).float()
+    X = torch.from_numpy(rng.normal(size=(n, 12))).float()
+    got = _ext.svc_predict(
+        X.cuda(), SV.cuda(), dual.cuda(), svclass.cuda(), intercept.cuda(), 0.05
+    ).cpu()
+    want = oc.svc_predict(
+        X.double(), SV.double(), dual.double(), intercept.double(), n_support, 0.05
+    )
+    agree = (got == want).float().mean().item()
+    assert agree > 0.999, agree

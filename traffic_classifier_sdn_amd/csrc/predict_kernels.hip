@@ -442,9 +442,14 @@ __global__ void svc_predict_kernel(const float* __restrict__ X,
     long long row = base + threadIdx.x;
     Row12 x;
     if (row < n) x = load_row12(X, row);
-    float acc[C * CR];
+    // f64 running accumulators + per-tile f32 partials: a single f32 chain
+    // over tens of thousands of SIGNED near-unit dual terms (thousands of
+    // alphas at the C bound in a non-separable fit) loses the O(1) decision
+    // value to rounding (measured: 24% wrong votes at 35K SVs); a 128-term
+    // f32 partial folded into f64 per tile is exact to ~1e-5 at any nsv
+    double acc[C * CR];
 #pragma unroll
-    for (int i = 0; i < C * CR; ++i) acc[i] = 0.f;
+    for (int i = 0; i < C * CR; ++i) acc[i] = 0.0;
 
     for (int tile = 0; tile < nsv; tile += SVC_TILE) {
       int cnt = min(SVC_TILE, nsv - tile);
@@ -459,6 +464,9 @@ __global__ void svc_predict_kernel(const float* __restrict__ X,
         s_cls[i] = svclass[tile + i];
       __syncthreads();
       if (row < n) {
+        float accT[C * CR];
+#pragma unroll
+        for (int i = 0; i < C * CR; ++i) accT[i] = 0.f;
         for (int s = 0; s < cnt; ++s) {
           float d = 0.f;
 #pragma unroll
@@ -473,11 +481,13 @@ __global__ void svc_predict_kernel(const float* __restrict__ X,
             if (c == cc) {
 #pragma unroll
               for (int r = 0; r < CR; ++r)
-                acc[cc * CR + r] =
-                    fmaf(s_dual[r * SVC_TILE + s], kv, acc[cc * CR + r]);
+                accT[cc * CR + r] =
+                    fmaf(s_dual[r * SVC_TILE + s], kv, accT[cc * CR + r]);
             }
           }
         }
+#pragma unroll
+        for (int i = 0; i < C * CR; ++i) acc[i] += (double)accT[i];
       }
     }
     if (row >= n) continue;
@@ -490,8 +500,8 @@ __global__ void svc_predict_kernel(const float* __restrict__ X,
     for (int i = 0; i < C; ++i)
 #pragma unroll
       for (int j = i + 1; j < C; ++j, ++p) {
-        float dec = acc[i * CR + (j - 1)] + acc[j * CR + i] + s_b[p];
-        if (dec > 0.f) votes[i]++; else votes[j]++;
+        double dec = acc[i * CR + (j - 1)] + acc[j * CR + i] + (double)s_b[p];
+        if (dec > 0.0) votes[i]++; else votes[j]++;
       }
     int best = -1, bi = 0;
 #pragma unroll
@@ -523,9 +533,12 @@ __launch_bounds__(256) __global__ void svc_predict_wave_kernel(
                         (threadIdx.x >> 6);
   if (row >= n) return;
   Row12 x = load_row12(X, row);  // broadcast load, all lanes same row
-  float acc[C * CR];
+  // f64 per-lane accumulators: the 64-way split already conditions the sum,
+  // but signed bounded-alpha duals at large nsv still overwhelm f32 (same
+  // hazard as the tiled kernel — see comment there)
+  double acc[C * CR];
 #pragma unroll
-  for (int i = 0; i < C * CR; ++i) acc[i] = 0.f;
+  for (int i = 0; i < C * CR; ++i) acc[i] = 0.0;
   for (int s = lane; s < nsv; s += WAVE) {
     Row12 sv = load_row12(SV, s);
     float d = 0.f;
@@ -541,7 +554,7 @@ __launch_bounds__(256) __global__ void svc_predict_wave_kernel(
       if (c == cc) {
 #pragma unroll
         for (int r = 0; r < CR; ++r)
-          acc[cc * CR + r] = fmaf(dual[(long long)r * nsv + s], kv, acc[cc * CR + r]);
+          acc[cc * CR + r] += (double)(dual[(long long)r * nsv + s] * kv);
       }
     }
   }
@@ -557,8 +570,8 @@ __launch_bounds__(256) __global__ void svc_predict_wave_kernel(
     for (int i = 0; i < C; ++i)
 #pragma unroll
       for (int j = i + 1; j < C; ++j, ++p) {
-        float dec = acc[i * CR + (j - 1)] + acc[j * CR + i] + intercept[p];
-        if (dec > 0.f) votes[i]++; else votes[j]++;
+        double dec = acc[i * CR + (j - 1)] + acc[j * CR + i] + (double)intercept[p];
+        if (dec > 0.0) votes[i]++; else votes[j]++;
       }
     int best = -1, bi = 0;
 #pragma unroll

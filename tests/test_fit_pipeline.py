@@ -193,3 +193,26 @@ def test_synthesize_quake_rows_plausible():
     m = GaussianNB().fit(Xtr, ytr)
     acc = (m.predict(Xte) == yte).mean()
     assert acc > 0.97  # published 6-class GNB accuracy is 98.63 on real quake
+
+
+def test_fit_with_synth_quake_six_classes(tmp_path):
+    """--with-synth-quake fits on the true 6-class set (shipped 5 classes +
+    D-ITG Quake3 replay rows) and writes loadable checkpoints."""
+    import json
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd.fit",
+         "--algos", "gaussiannb", "--with-synth-quake",
+         "--out", str(tmp_path), "--json"],
+        capture_output=True, text=True, timeout=300,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    )
+    assert r.returncode == 0, r.stderr[-800:]
+    d = json.loads(r.stdout.strip().splitlines()[-1])
+    assert d["accuracy"] > 0.97
+    from traffic_classifier_sdn_amd.models import load_model
+
+    m = load_model(str(tmp_path / "GaussianNB.npz"))
+    assert sorted(str(c) for c in m.classes_) == ["dns", "game", "ping", "quake", "telnet", "voice"]

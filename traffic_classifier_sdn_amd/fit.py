@@ -140,6 +140,11 @@ def main(argv: Optional[Sequence[str]] = None) -> int:
     ap.add_argument("--seed", type=int, default=101)
     ap.add_argument("--sklearn-pickles", action="store_true", help="also write sklearn-1.0.1-layout pickles")
     ap.add_argument("--json", action="store_true", help="machine-readable result lines")
+    ap.add_argument("--with-synth-quake", action="store_true",
+                    help="add synthetic quake rows (D-ITG Quake3 replay) for a "
+                         "true 6-class fit — the reference's quake CSV is not "
+                         "shipped; accuracies are then against the synthetic "
+                         "6-class split, not the published 5-class-real rows")
     args = ap.parse_args(argv)
 
     # torchrun-aware: initialize the group if launched with a WORLD_SIZE
@@ -154,7 +159,12 @@ def main(argv: Optional[Sequence[str]] = None) -> int:
     sharded = dist.is_initialized() and dist.world_size() > 1
     rank = dist.rank() if dist.is_initialized() else 0
 
-    X, y = load_reference_dataset(data_dir=args.data_dir)
+    if args.with_synth_quake:
+        from .utils.datasets import load_six_class_dataset
+
+        X, y = load_six_class_dataset(data_dir=args.data_dir)
+    else:
+        X, y = load_reference_dataset(data_dir=args.data_dir)
     Xtr, Xte, ytr, yte = train_test_split_ref(X, y, test_size=args.test_size, random_state=args.seed)
 
     algos = list(ALGOS) if args.algos == "all" else [a.strip().lower() for a in args.algos.split(",")]
@@ -179,7 +189,9 @@ def main(argv: Optional[Sequence[str]] = None) -> int:
                     print(f"WARNING: {algo}: {e}", file=sys.stderr)
             res["checkpoint"] = npz_path
             pub = res.get("published_accuracy")
-            if pub is not None and res["accuracy"] < pub - 0.02:
+            # published numbers were measured on the real 6-class rows; with
+            # a synthetic 6th class the comparison is informational only
+            if pub is not None and res["accuracy"] < pub - 0.02 and not args.with_synth_quake:
                 ok = False
             if args.json:
                 print(json.dumps({k: v for k, v in res.items() if k != "confusion_matrix"}))

@@ -71,7 +71,7 @@ DEV float km_dist2(const float* __restrict__ q, const float* __restrict__ r) {
   return a + b;
 }
 
-__launch_bounds__(256) __global__ void knn_mfma_kernel(
+__launch_bounds__(256, 4) __global__ void knn_mfma_kernel(
     const float* __restrict__ Q, const float* __restrict__ R,
     const float* __restrict__ cmean,  // [12] reference column means
     float* __restrict__ part_d,       // [S, nq, k] refined exact d^2
@@ -80,6 +80,7 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
   __shared__ float s_qt[KM_F][KM_QB];            // centered Q^T
   __shared__ float s_rt[2][KM_F][KM_TB];         // centered R-tile^T (2 bufs)
   __shared__ float s_rn[2][KM_TB];               // ||r-c||^2 (FLT_MAX pad)
+  __shared__ float s_cm[KM_F];                   // column means (staging only)
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -102,16 +103,18 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
     return;
   }
 
-  float cm[KM_F];
-#pragma unroll
-  for (int j = 0; j < KM_F; ++j) cm[j] = cmean[j];
+  // column means live in LDS, not registers: they are read only in the
+  // staging phases, and the 12 VGPRs they'd occupy are part of the margin
+  // that keeps this kernel at 4 waves/SIMD
+  if (tid < KM_F) s_cm[tid] = cmean[tid];
+  __syncthreads();
 
   // ---- stage the query block: centered, transposed -----------------------
   for (int i = tid; i < KM_QB; i += blockDim.x) {
     long long q = min(qb0 + i, nq - 1);  // clamp tail (outputs masked later)
     Row12 x = load_row12(Q, q);
 #pragma unroll
-    for (int j = 0; j < KM_F; ++j) s_qt[j][i] = x.v[j] - cm[j];
+    for (int j = 0; j < KM_F; ++j) s_qt[j][i] = x.v[j] - s_cm[j];
   }
   __syncthreads();
 
@@ -171,7 +174,7 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
       float rn = 0.f;
 #pragma unroll
       for (int j = 0; j < KM_F; ++j) {
-        float rc = pf_row.v[j] - cm[j];
+        float rc = pf_row.v[j] - s_cm[j];
         s_rt[buf][j][pf_i] = rc;
         rn = fmaf(rc, rc, rn);
       }
@@ -205,10 +208,14 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
       // latency per candidate check — 32 dependent waits per subtile
       const float4* rn4 =
           reinterpret_cast<const float4*>(rn_t + ct * 32 + 4 * half);
-      float4 rnv[4];
+      float4 rnv[2];
 #pragma unroll
-      for (int r4 = 0; r4 < 4; ++r4) rnv[r4] = rn4[2 * r4];
-      const float* rnf = reinterpret_cast<const float*>(rnv);
+      for (int r4 = 0; r4 < 2; ++r4) rnv[r4] = rn4[2 * r4];
+      float4 rnw[2];
+#pragma unroll
+      for (int r4 = 0; r4 < 2; ++r4) rnw[r4] = rn4[4 + 2 * r4];
+      const float* rnf0 = reinterpret_cast<const float*>(rnv);
+      const float* rnf1 = reinterpret_cast<const float*>(rnw);
 #pragma unroll
       for (int qti = 0; qti < KM_NQT; ++qti) {
         f32x16 acc = {};
@@ -222,16 +229,20 @@ __launch_bounds__(256) __global__ void knn_mfma_kernel(
         // per-candidate skip branches: the common (no-survivor) path is 16
         // fma + 15 min + 1 compare, all VALU, and the insert block's exec
         // mask is wave-collective (taken only when SOME lane has a winner).
-        float keyv[16];
+        // Keys are NOT kept in registers — the rare insert path recomputes
+        // them from acc/rnf, saving 16 VGPRs (the difference between 3 and
+        // 4 waves/SIMD at this kernel's register budget).
+        float kmin = fmaf(-2.f, acc[0], rnf0[0]);
 #pragma unroll
-        for (int g = 0; g < 16; ++g) keyv[g] = fmaf(-2.f, acc[g], rnf[g]);
-        float kmin = keyv[0];
+        for (int g = 1; g < 8; ++g)
+          kmin = fminf(kmin, fmaf(-2.f, acc[g], rnf0[g]));
 #pragma unroll
-        for (int g = 1; g < 16; ++g) kmin = fminf(kmin, keyv[g]);
+        for (int g = 8; g < 16; ++g)
+          kmin = fminf(kmin, fmaf(-2.f, acc[g], rnf1[g - 8]));
         if (kmin < wkey[qti]) {
 #pragma unroll
           for (int g = 0; g < 16; ++g) {
-            float key = keyv[g];
+            float key = fmaf(-2.f, acc[g], g < 8 ? rnf0[g] : rnf1[g - 8]);
             if (key < wkey[qti]) {
               const int cand = ct * 32 + km_rowmap(g, half);
               int ws = 0;

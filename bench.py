@@ -161,7 +161,8 @@ def bench_knn(args, rank, world, device, use_gpu):
     X_real, _ = load_reference_dataset()
     Xref = synthetic_flow_rows(ref_rows, seed=args.seed + 7000 + 1000 * rank, reference_X=X_real)
     yref = np.random.default_rng(args.seed + rank).integers(0, 6, size=ref_rows)
-    m = KNeighborsClassifier(n_neighbors=5, batch_rows=args.knn_queries, device=device)
+    m = KNeighborsClassifier(n_neighbors=5, batch_rows=args.knn_queries,
+                             device=device, approx=args.knn_approx)
     m.fit(Xref, yref, sharded=world > 1)
     Q = torch.from_numpy(
         synthetic_flow_rows(args.knn_queries, seed=args.seed + 31 + rank, reference_X=X_real)
@@ -187,6 +188,7 @@ def bench_knn(args, rank, world, device, use_gpu):
             "parallelism": f"shard{world}",
             "reference_rows_total": ref_rows * world,
             "reference_rows_per_gpu": ref_rows,
+            "selection": "bf16-coarse+exact-refine" if args.knn_approx else "exact-f32",
         },
     )
 
@@ -465,6 +467,9 @@ def main() -> int:
     ap.add_argument("--rows-per-gpu", type=int, default=10_000_000)
     ap.add_argument("--knn-ref-rows-per-gpu", type=int, default=12_500_000)
     ap.add_argument("--knn-queries", type=int, default=65_536)
+    ap.add_argument("--knn-approx", action="store_true",
+                    help="bf16 coarse-pass KNN selection (exact-f32 refine; "
+                         "measured recall, opt-in — default is exact)")
     ap.add_argument("--svc-rows", type=int, default=1_000_000)
     ap.add_argument("--svc-iters-per-step", type=int, default=200)
     ap.add_argument("--svc-full-max-iter", type=int, default=20_000)

@@ -388,3 +388,39 @@ def test_svc_predict_large_nsv_accumulation():
     )
     agree = (got == want).float().mean().item()
     assert agree > 0.999, agree
+
+
+@pytest.mark.gpu
+def test_knn_bf16_coarse_pass_recall():
+    """Opt-in bf16 coarse-pass KNN (approx=1): candidate pool ranked by the
+    bf16 key, every pooled candidate refined with the exact f32 distance.
+    Recall vs the exact kernel must be >= 0.999 @ k=5 and the returned
+    distances must be the exact f32 values of the returned neighbours."""
+    from traffic_classifier_sdn_amd.ops.gpu import _ext, _knn_cmean
+    from traffic_classifier_sdn_amd.utils.datasets import synthetic_flow_rows
+
+    R = torch.from_numpy(synthetic_flow_rows(400_000, seed=3)).float().cuda().contiguous()
+    Q = torch.from_numpy(synthetic_flow_rows(4096, seed=4)).float().cuda().contiguous()
+    k = 5
+    d_ex, i_ex = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), None, k, 0, 0, 8, 0)
+    d_ap, i_ap = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), None, k, 0, 0, 8, 1)
+    ex = i_ex.cpu().numpy()
+    ap = i_ap.cpu().numpy()
+    recall = np.mean([len(set(ex[i]) & set(ap[i])) / k for i in range(ex.shape[0])])
+    assert recall >= 0.999, recall
+    # distances of the approx result are the EXACT f32 values (refined)
+    Qc, Rc = Q.cpu(), R.cpu()
+    sub = np.random.default_rng(0).choice(4096, 200, replace=False)
+    for qi in sub:
+        for j in range(k):
+            ri = int(ap[qi, j])
+            want = float(((Qc[qi] - Rc[ri]) ** 2).sum())
+            got = float(d_ap[qi, j])
+            assert abs(got - want) <= 1e-4 * max(1.0, want), (qi, j)
+
+    # fused vote agreement with the exact path
+    y = torch.from_numpy(np.random.default_rng(0).integers(0, 6, size=400_000)).cuda()
+    y8 = y.to(torch.uint8).contiguous()
+    _, _, lab_ex = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), y8, k, 6, 0, 8, 0)
+    _, _, lab_ap = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), y8, k, 6, 0, 8, 1)
+    assert (lab_ex.cpu() == lab_ap.cpu()).float().mean().item() > 0.995

@@ -11,13 +11,15 @@ R = torch.rand(n_ref, 12, device="cuda", generator=g) * 1e5
 Q = torch.rand(n_q, 12, device="cuda", generator=g) * 1e5
 torch.cuda.synchronize()
 print(f"resident reference: {R.numel()*4/2**30:.1f} GiB on", torch.cuda.get_device_name(0))
-d, i = og.knn_topk(Q, R, k)  # warmup
-torch.cuda.synchronize()
-t0 = time.perf_counter()
-d, i = og.knn_topk(Q, R, k)
-torch.cuda.synchronize()
-dt = time.perf_counter() - t0
-print(f"top-{k} of {n_q} queries vs {n_ref/1e9:.0f}B rows: {dt:.2f}s "
-      f"= {n_q*n_ref/dt:.3g} candidate distances/s")
-assert int(i.max()) < n_ref and int(i.min()) >= 0
+for approx in (False, True):
+    d, i = og.knn_topk(Q, R, k, approx=approx)  # warmup
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    d, i = og.knn_topk(Q, R, k, approx=approx)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    tag = "bf16-coarse+exact-refine" if approx else "exact-f32"
+    print(f"[{tag}] top-{k} of {n_q} queries vs {n_ref/1e9:.0f}B rows: "
+          f"{dt:.2f}s = {n_q*n_ref/dt:.3g} candidate distances/s")
+    assert int(i.max()) < n_ref and int(i.min()) >= 0
 print("OK")

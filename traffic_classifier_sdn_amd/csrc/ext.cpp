@@ -37,7 +37,7 @@ void launch_knn_topk(const float*, const float*, const unsigned char*, float*,
                      hipStream_t);
 void launch_knn_mfma(const float*, const float*, const float*,
                      const unsigned char*, float*, int*, float*, int*, int*,
-                     long long, long long, int, int, int, long long,
+                     long long, long long, int, int, int, long long, int,
                      hipStream_t);
 void launch_gnb_fit_stats(const double*, const long long*, double*, double*,
                           double*, long long, int, hipStream_t);
@@ -187,7 +187,8 @@ static std::vector<torch::Tensor> knn_topk_mfma(torch::Tensor Q,
                                                 c10::optional<torch::Tensor> ry,
                                                 int64_t k, int64_t C,
                                                 int64_t idx_base,
-                                                int64_t n_shards) {
+                                                int64_t n_shards,
+                                                int64_t approx) {
   CHECK_IN(Q, torch::kFloat32);
   CHECK_IN(R, torch::kFloat32);
   CHECK_IN(cmean, torch::kFloat32);
@@ -216,7 +217,7 @@ static std::vector<torch::Tensor> knn_topk_mfma(torch::Tensor Q,
                   cmean.data_ptr<float>(), ry_ptr, part_d.data_ptr<float>(),
                   part_i.data_ptr<int>(), dist.data_ptr<float>(),
                   idx.data_ptr<int>(), lab_ptr, nq, nr, S, (int)k, (int)C,
-                  idx_base, cur_stream());
+                  idx_base, (int)approx, cur_stream());
   if (ry.has_value()) return {dist, idx, lab};
   return {dist, idx};
 }
@@ -349,7 +350,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_predict", &rf_predict, "packed-forest traversal + vote");
   m.def("svc_predict", &svc_predict, "RBF Gram + OVO vote");
   m.def("knn_topk", &knn_topk, "brute-force top-k (+fused vote)");
-  m.def("knn_topk_mfma", &knn_topk_mfma, "MFMA distance-GEMM top-k (+fused vote)");
+  m.def("knn_topk_mfma", &knn_topk_mfma, "MFMA distance-GEMM top-k (+fused vote)",
+        py::arg("Q"), py::arg("R"), py::arg("cmean"), py::arg("ry"),
+        py::arg("k"), py::arg("C"), py::arg("idx_base"), py::arg("n_shards"),
+        py::arg("approx") = 0);
   m.def("gnb_fit_stats", &gnb_fit_stats, "per-class sufficient stats");
   m.def("logistic_grad", &logistic_grad, "fused CE loss+grad");
   m.def("flow_features", &flow_features, "counters -> 12 features");

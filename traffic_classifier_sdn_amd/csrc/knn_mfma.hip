@@ -342,6 +342,231 @@ __launch_bounds__(256, 4) __global__ void knn_mfma_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// bf16 coarse-pass variant (OPT-IN approximate mode, `approx=True`).
+//
+// Same swapped-operand structure, but the distance GEMM runs on
+// v_mfma_f32_32x32x16_bf16 — 16x the f32-MFMA rate — over bf16-rounded
+// centered rows (K=16: the 12 features + 4 zero pads, ONE MFMA per 32x32
+// subtile instead of six).  Selection keeps the FULL KM_KMAX=8 candidates
+// per lane half (16 per query — a >=2x safety margin over k<=8) ranked by
+// the approximate bf16 key; the tail then refines EVERY pooled candidate
+// with the exact f32 direct-difference distance and emits the exact-best k
+// of the pool.  The output is exact distances over an approximate candidate
+// pool: recall is measured, not proven (tests pin >=0.999 @ k=5 on
+// flow-feature-scale data) — the default path stays the exact f32 kernel.
+// A/B fragment layout hardware-verified: tools/dbg_mfma_bf16_probe.hip.
+// ---------------------------------------------------------------------------
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+__launch_bounds__(256, 4) __global__ void knn_mfma_bf16_kernel(
+    const float* __restrict__ Q, const float* __restrict__ R,
+    const float* __restrict__ cmean, float* __restrict__ part_d,
+    int* __restrict__ part_i, long long nq, long long nr, int k,
+    long long shard_rows) {
+  // per-candidate-contiguous bf16 layout: one lane's 8-element K-chunk is a
+  // single 16-byte ds_read (s_qt/s_rt rows are 32 B)
+  __shared__ __attribute__((aligned(16))) __bf16 s_qt[KM_QB][16];
+  __shared__ __attribute__((aligned(16))) __bf16 s_rt[2][KM_TB][16];
+  __shared__ float s_rn[2][KM_TB];
+  __shared__ float s_cm[KM_F];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wave = tid >> 6;
+  const int half = lane >> 5;
+  const int l31 = lane & 31;
+  const long long qb0 = (long long)blockIdx.x * KM_QB;
+  const int shard = blockIdx.y;
+  const long long r0 = (long long)shard * shard_rows;
+  const long long r1 = min(nr, r0 + shard_rows);
+  if (r0 >= r1) {
+    for (int row = tid; row < KM_QB; row += blockDim.x) {
+      long long q = qb0 + row;
+      if (q < nq)
+        for (int j = 0; j < k; ++j) {
+          part_d[((long long)shard * nq + q) * k + j] = FLT_MAX;
+          part_i[((long long)shard * nq + q) * k + j] = -1;
+        }
+    }
+    return;
+  }
+
+  if (tid < KM_F) s_cm[tid] = cmean[tid];
+  __syncthreads();
+
+  for (int i = tid; i < KM_QB; i += blockDim.x) {
+    long long q = min(qb0 + i, nq - 1);
+    Row12 x = load_row12(Q, q);
+#pragma unroll
+    for (int j = 0; j < KM_F; ++j) s_qt[i][j] = (__bf16)(x.v[j] - s_cm[j]);
+#pragma unroll
+    for (int j = KM_F; j < 16; ++j) s_qt[i][j] = (__bf16)0.f;
+  }
+  __syncthreads();
+
+  bf16x8 bfrag[KM_NQT];
+#pragma unroll
+  for (int qti = 0; qti < KM_NQT; ++qti)
+    bfrag[qti] = *reinterpret_cast<const bf16x8*>(
+        &s_qt[(wave + 4 * qti) * 32 + l31][half * 8]);
+
+  float lk[KM_NQT][KM_KMAX];
+  int li[KM_NQT][KM_KMAX];
+  float wkey[KM_NQT];
+#pragma unroll
+  for (int t = 0; t < KM_NQT; ++t) {
+#pragma unroll
+    for (int j = 0; j < KM_KMAX; ++j) {
+      lk[t][j] = FLT_MAX;
+      li[t][j] = -1;
+    }
+    wkey[t] = FLT_MAX;
+  }
+
+  const int pf_i = tid;
+  Row12 pf_row;
+  bool pf_valid = false;
+  if (r0 + pf_i < r1) {
+    pf_row = load_row12(R, r0 + pf_i);
+    pf_valid = true;
+  }
+  int t = 0;
+  for (long long tb = r0; tb < r1; tb += KM_TB, ++t) {
+    const int buf = t & 1;
+    if (pf_valid) {
+      // rn from the bf16-ROUNDED values: keys then rank the rounded
+      // geometry consistently (the refine pass restores exact f32)
+      float rn = 0.f;
+#pragma unroll
+      for (int j = 0; j < KM_F; ++j) {
+        __bf16 rb = (__bf16)(pf_row.v[j] - s_cm[j]);
+        s_rt[buf][pf_i][j] = rb;
+        float rc = (float)rb;
+        rn = fmaf(rc, rc, rn);
+      }
+#pragma unroll
+      for (int j = KM_F; j < 16; ++j) s_rt[buf][pf_i][j] = (__bf16)0.f;
+      s_rn[buf][pf_i] = rn;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) s_rt[buf][pf_i][j] = (__bf16)0.f;
+      s_rn[buf][pf_i] = FLT_MAX;
+    }
+    {
+      long long nxt = tb + KM_TB + pf_i;
+      pf_valid = nxt < r1;
+      if (pf_valid) pf_row = load_row12(R, nxt);
+    }
+    __syncthreads();
+
+    const __bf16(*rt)[16] = s_rt[buf];
+    const float* rn_t = s_rn[buf];
+
+    for (int ct = 0; ct < KM_TB / 32; ++ct) {
+      bf16x8 afrag =
+          *reinterpret_cast<const bf16x8*>(&rt[ct * 32 + l31][half * 8]);
+      const float4* rn4 =
+          reinterpret_cast<const float4*>(rn_t + ct * 32 + 4 * half);
+      float4 rnv[2];
+#pragma unroll
+      for (int r4 = 0; r4 < 2; ++r4) rnv[r4] = rn4[2 * r4];
+      float4 rnw[2];
+#pragma unroll
+      for (int r4 = 0; r4 < 2; ++r4) rnw[r4] = rn4[4 + 2 * r4];
+      const float* rnf0 = reinterpret_cast<const float*>(rnv);
+      const float* rnf1 = reinterpret_cast<const float*>(rnw);
+#pragma unroll
+      for (int qti = 0; qti < KM_NQT; ++qti) {
+        f32x16 acc = {};
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, bfrag[qti], acc,
+                                                      0, 0, 0);
+        float kmin = fmaf(-2.f, acc[0], rnf0[0]);
+#pragma unroll
+        for (int g = 1; g < 8; ++g)
+          kmin = fminf(kmin, fmaf(-2.f, acc[g], rnf0[g]));
+#pragma unroll
+        for (int g = 8; g < 16; ++g)
+          kmin = fminf(kmin, fmaf(-2.f, acc[g], rnf1[g - 8]));
+        if (kmin < wkey[qti]) {
+#pragma unroll
+          for (int g = 0; g < 16; ++g) {
+            float key = fmaf(-2.f, acc[g], g < 8 ? rnf0[g] : rnf1[g - 8]);
+            if (key < wkey[qti]) {
+              const int cand = ct * 32 + km_rowmap(g, half);
+              // pool is the FULL KM_KMAX regardless of runtime k: the
+              // surplus is the refine stage's safety margin
+              int ws = 0;
+              float wv = -FLT_MAX;
+#pragma unroll
+              for (int j = 0; j < KM_KMAX; ++j)
+                if (lk[qti][j] > wv) {
+                  wv = lk[qti][j];
+                  ws = j;
+                }
+              lk[qti][ws] = key;
+              li[qti][ws] = (int)(tb + cand);
+              wv = -FLT_MAX;
+#pragma unroll
+              for (int j = 0; j < KM_KMAX; ++j)
+                if (lk[qti][j] > wv) wv = lk[qti][j];
+              wkey[qti] = wv;
+            }
+          }
+        }
+      }
+    }
+  }
+  __syncthreads();
+
+  // tail: refine EVERY pooled candidate exactly, emit the exact-best k
+#pragma unroll
+  for (int qti = 0; qti < KM_NQT; ++qti) {
+    const int qt = wave + 4 * qti;
+    const long long q = qb0 + qt * 32 + l31;
+    int oi[2 * KM_KMAX];
+#pragma unroll
+    for (int j = 0; j < KM_KMAX; ++j) {
+      oi[j] = li[qti][j];
+      oi[KM_KMAX + j] = __shfl(li[qti][j], lane ^ 32, WAVE);
+    }
+    if (half == 0 && q < nq) {
+      Row12 xq = load_row12(Q, q);
+      float dk[KM_KMAX];
+      int ik[KM_KMAX];
+#pragma unroll
+      for (int j = 0; j < KM_KMAX; ++j) {
+        dk[j] = FLT_MAX;
+        ik[j] = -1;
+      }
+      for (int j = 0; j < 2 * KM_KMAX; ++j) {
+        if (oi[j] < 0) continue;
+        Row12 xr = load_row12(R, oi[j]);
+        float dv = km_dist2(xq.v, xr.v);
+        int iv = oi[j];
+        if (dv > dk[k - 1] ||
+            (dv == dk[k - 1] && ik[k - 1] >= 0 && (unsigned)iv > (unsigned)ik[k - 1]))
+          continue;
+        int b = k - 1;
+        while (b > 0 && (dk[b - 1] > dv ||
+                         (dk[b - 1] == dv && (unsigned)ik[b - 1] > (unsigned)iv))) {
+          dk[b] = dk[b - 1];
+          ik[b] = ik[b - 1];
+          --b;
+        }
+        dk[b] = dv;
+        ik[b] = iv;
+      }
+      long long o = ((long long)shard * nq + q) * k;
+      for (int j = 0; j < k; ++j) {
+        part_d[o + j] = dk[j];
+        part_i[o + j] = ik[j];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Merge the S per-shard lists of one query into the final top-k (+ fused
 // uniform vote when labels are supplied).  One thread per query.
 // ---------------------------------------------------------------------------
@@ -404,11 +629,15 @@ extern "C" void launch_knn_mfma(const float* Q, const float* R, const float* cme
                                 const unsigned char* ry, float* part_d, int* part_i,
                                 float* out_d, int* out_i, int* out_lab,
                                 long long nq, long long nr, int S, int k, int C,
-                                long long idx_base, hipStream_t stream) {
+                                long long idx_base, int approx, hipStream_t stream) {
   long long shard_rows = (nr + S - 1) / S;
   dim3 grid((unsigned)((nq + KM_QB - 1) / KM_QB), (unsigned)S);
-  hipLaunchKernelGGL(knn_mfma_kernel, grid, dim3(256), 0, stream, Q, R, cmean,
-                     part_d, part_i, nq, nr, k, shard_rows);
+  if (approx)
+    hipLaunchKernelGGL(knn_mfma_bf16_kernel, grid, dim3(256), 0, stream, Q, R,
+                       cmean, part_d, part_i, nq, nr, k, shard_rows);
+  else
+    hipLaunchKernelGGL(knn_mfma_kernel, grid, dim3(256), 0, stream, Q, R, cmean,
+                       part_d, part_i, nq, nr, k, shard_rows);
   int block = 256;
   int mgrid = (int)((nq + block - 1) / block);
   hipLaunchKernelGGL(knn_merge_kernel, dim3(mgrid), dim3(block), 0, stream,

@@ -22,10 +22,15 @@ from .base import ArrayLike, Estimator, as_tensor, encode_labels
 class KNeighborsClassifier(Estimator):
     kind = "kneighbors"
 
-    def __init__(self, n_neighbors: int = 5, batch_rows: int = 65536, device: Optional[str] = None):
+    def __init__(self, n_neighbors: int = 5, batch_rows: int = 65536,
+                 device: Optional[str] = None, approx: bool = False):
         super().__init__(device)
         self.n_neighbors = n_neighbors
         self.batch_rows = batch_rows
+        # opt-in bf16 coarse-pass selection on GPU (exact-f32 refine over a
+        # 16-candidate pool; measured recall >= 0.999 @ k=5 on flow-feature
+        # scale data — csrc/knn_mfma.hip).  The default is the exact kernel.
+        self.approx = approx
         self.fit_X_: Optional[torch.Tensor] = None
         self.y_: Optional[torch.Tensor] = None
         self.sharded_ = False
@@ -52,7 +57,8 @@ class KNeighborsClassifier(Estimator):
         k = min(self.n_neighbors, self.fit_X_.shape[0])
         dists, idxs = [], []
         for lo in range(0, Xt.shape[0], self.batch_rows):
-            d, i = ops.knn_topk(Xt[lo : lo + self.batch_rows], self.fit_X_, k)
+            d, i = ops.knn_topk(Xt[lo : lo + self.batch_rows], self.fit_X_, k,
+                                approx=self.approx)
             dists.append(d)
             idxs.append(i)
         return torch.cat(dists), torch.cat(idxs)

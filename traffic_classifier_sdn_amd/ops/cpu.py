@@ -129,7 +129,8 @@ def kmeans_assign(
 # ----------------------------------------------------------------------
 
 
-def knn_topk(Q: torch.Tensor, R: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor]:
+def knn_topk(Q: torch.Tensor, R: torch.Tensor, k: int, approx: bool = False) -> Tuple[torch.Tensor, torch.Tensor]:
+    # approx is a GPU-path option (bf16 coarse pass); the CPU oracle is exact
     """Brute-force k smallest squared distances. Returns (dist[nq,k], idx[nq,k]),
     sorted ascending (ties by lower index, matching sklearn's ordering)."""
     d = pairwise_sqdist(Q, R)

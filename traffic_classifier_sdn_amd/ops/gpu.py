@@ -135,28 +135,37 @@ def _knn_shards(nr: int) -> int:
     return max(1, min(64, nr // 262_144))
 
 
-def knn_topk(Q: torch.Tensor, R: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor]:
+def knn_topk(Q: torch.Tensor, R: torch.Tensor, k: int, approx: bool = False) -> Tuple[torch.Tensor, torch.Tensor]:
     if Q.shape[1] != 12:
         return _cpu.knn_topk(Q, R, k)
     Qf, Rf = _f32(Q), _f32(R)
-    if R.shape[0] >= _KNN_MFMA_MIN_ROWS and k <= 8:
-        dist, idx = _ext.knn_topk_mfma(Qf, Rf, _knn_cmean(Rf), None, k, 0, 0, _knn_shards(R.shape[0]))
+    if (approx or R.shape[0] >= _KNN_MFMA_MIN_ROWS) and k <= 8:
+        dist, idx = _ext.knn_topk_mfma(
+            Qf, Rf, _knn_cmean(Rf), None, k, 0, 0, _knn_shards(R.shape[0]),
+            1 if approx else 0,
+        )
         return dist, idx
     dist, idx = _ext.knn_topk(Qf, Rf, None, k, 0, 0)
     return dist, idx
 
 
 def knn_classify(
-    Q: torch.Tensor, R: torch.Tensor, y: torch.Tensor, k: int, n_classes: int, idx_base: int = 0
+    Q: torch.Tensor, R: torch.Tensor, y: torch.Tensor, k: int, n_classes: int,
+    idx_base: int = 0, approx: bool = False
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
-    """Fused top-k + uniform vote (labels also returned for sharded merge)."""
+    """Fused top-k + uniform vote (labels also returned for sharded merge).
+
+    ``approx=True`` selects the bf16 coarse-pass kernel: 16x the f32 MFMA
+    rate, exact-f32 refine over a 16-candidate pool per query (measured
+    recall, not proven — see csrc/knn_mfma.hip)."""
     if Q.shape[1] != 12:
         return _cpu.knn_classify(Q, R, y, k, n_classes, idx_base)
     y8 = y.to(torch.uint8).contiguous()
     Qf, Rf = _f32(Q), _f32(R)
-    if R.shape[0] >= _KNN_MFMA_MIN_ROWS and k <= 8 and n_classes <= 16:
+    if (approx or R.shape[0] >= _KNN_MFMA_MIN_ROWS) and k <= 8 and n_classes <= 16:
         dist, idx, lab = _ext.knn_topk_mfma(
-            Qf, Rf, _knn_cmean(Rf), y8, k, n_classes, idx_base, _knn_shards(R.shape[0])
+            Qf, Rf, _knn_cmean(Rf), y8, k, n_classes, idx_base,
+            _knn_shards(R.shape[0]), 1 if approx else 0
         )
         return dist, idx, lab
     dist, idx, lab = _ext.knn_topk(Qf, Rf, y8, k, n_classes, idx_base)

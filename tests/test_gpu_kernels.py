@@ -424,3 +424,34 @@ def test_knn_bf16_coarse_pass_recall():
     _, _, lab_ex = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), y8, k, 6, 0, 8, 0)
     _, _, lab_ap = _ext.knn_topk_mfma(Q, R, _knn_cmean(R), y8, k, 6, 0, 8, 1)
     assert (lab_ex.cpu() == lab_ap.cpu()).float().mean().item() > 0.995
+
+
+@pytest.mark.gpu
+def test_smo_wss2_converges():
+    """The WSS-2 fused pipeline (measured off by default — see
+    profiles/svc_wss2_r02.md) reaches the same optimum as WSS-1."""
+    from traffic_classifier_sdn_amd.models.svc_fit import (
+        _smo_fused_gpu,
+        _smo_intercept,
+    )
+
+    rng = np.random.default_rng(9)
+    Xn = (rng.normal(size=(3000, 12)) * 2).astype(np.float32)
+    yn = np.where(Xn[:, 0] + 0.5 * Xn[:, 3] + rng.normal(size=3000) * 0.5 > 0, 1.0, -1.0)
+    X = torch.from_numpy(Xn).cuda()
+    yv = torch.from_numpy(yn.astype(np.float32)).cuda()
+    res = {}
+    for wss2 in (False, True):
+        alpha = torch.zeros(3000, dtype=torch.float64, device="cuda")
+        grad = -torch.ones(3000, dtype=torch.float64, device="cuda")
+        it = _smo_fused_gpu(X, yv, alpha, grad, 1.0, 0.05, 1e-3, 20000, wss2=wss2)
+        b = _smo_intercept(yv, alpha, grad, 1.0, "cuda")
+        res[wss2] = (alpha.clone(), b, it)
+    a1, b1, _ = res[False]
+    a2, b2, _ = res[True]
+    assert b2 == pytest.approx(b1, abs=5e-2)
+    # same optimum: decision agreement on the training rows
+    K = torch.exp(-0.05 * torch.cdist(X, X) ** 2).double()
+    d1 = K @ (a1 * yv.double()) + b1
+    d2 = K @ (a2 * yv.double()) + b2
+    assert float(((d1 > 0) == (d2 > 0)).float().mean()) > 0.995

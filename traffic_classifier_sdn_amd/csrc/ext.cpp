@@ -55,6 +55,17 @@ void launch_smo_solve(const float*, const float*, double*, const double*,
                       float, hipStream_t);
 void launch_smo_update_dev(const float*, const float*, double*, const float*,
                            const double*, float, long long, hipStream_t);
+void launch_smo_row(const float*, const unsigned long long*, const double*,
+                    float*, float, long long, hipStream_t);
+void launch_smo_select2(const float*, const double*, const double*,
+                        const float*, unsigned long long*, const double*,
+                        double, long long, hipStream_t);
+void launch_smo_solve2(const float*, const float*, double*, const double*,
+                       unsigned long long*, float*, double*, double, double,
+                       float, hipStream_t);
+void launch_smo_update_dev2(const float*, const float*, double*, const float*,
+                            const double*, const float*, float, long long,
+                            hipStream_t);
 void launch_smo_update(const float*, const float*, double*, const float*,
                        double, double, float, long long, hipStream_t);
 }
@@ -326,6 +337,66 @@ static void smo_update_dev(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
                         cur_stream());
 }
 
+static void smo_row(torch::Tensor X, torch::Tensor sel, torch::Tensor sol,
+                    torch::Tensor krow, double gamma) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(sel, torch::kInt64);
+  CHECK_IN(sol, torch::kFloat64);
+  CHECK_IN(krow, torch::kFloat32);
+  launch_smo_row(X.data_ptr<float>(),
+                 reinterpret_cast<unsigned long long*>(sel.data_ptr<int64_t>()),
+                 sol.data_ptr<double>(), krow.data_ptr<float>(), (float)gamma,
+                 X.size(0), cur_stream());
+}
+
+static void smo_select2(torch::Tensor y, torch::Tensor alpha,
+                        torch::Tensor grad, torch::Tensor krow,
+                        torch::Tensor sel, torch::Tensor sol, double C) {
+  CHECK_IN(y, torch::kFloat32);
+  CHECK_IN(alpha, torch::kFloat64);
+  CHECK_IN(grad, torch::kFloat64);
+  CHECK_IN(krow, torch::kFloat32);
+  CHECK_IN(sel, torch::kInt64);
+  CHECK_IN(sol, torch::kFloat64);
+  launch_smo_select2(y.data_ptr<float>(), alpha.data_ptr<double>(),
+                     grad.data_ptr<double>(), krow.data_ptr<float>(),
+                     reinterpret_cast<unsigned long long*>(sel.data_ptr<int64_t>()),
+                     sol.data_ptr<double>(), C, y.size(0), cur_stream());
+}
+
+static void smo_solve2(torch::Tensor X, torch::Tensor y, torch::Tensor alpha,
+                       torch::Tensor grad, torch::Tensor sel, torch::Tensor rows,
+                       torch::Tensor sol, double C, double tol, double gamma) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(y, torch::kFloat32);
+  CHECK_IN(alpha, torch::kFloat64);
+  CHECK_IN(grad, torch::kFloat64);
+  CHECK_IN(sel, torch::kInt64);
+  CHECK_IN(rows, torch::kFloat32);
+  CHECK_IN(sol, torch::kFloat64);
+  TORCH_CHECK(sel.numel() >= 3, "WSS-2 sel buffer must be u64[3]");
+  launch_smo_solve2(X.data_ptr<float>(), y.data_ptr<float>(),
+                    alpha.data_ptr<double>(), grad.data_ptr<double>(),
+                    reinterpret_cast<unsigned long long*>(sel.data_ptr<int64_t>()),
+                    rows.data_ptr<float>(), sol.data_ptr<double>(), C, tol,
+                    (float)gamma, cur_stream());
+}
+
+static void smo_update_dev2(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
+                            torch::Tensor rows, torch::Tensor sol,
+                            torch::Tensor krow, double gamma) {
+  CHECK_IN(X, torch::kFloat32);
+  CHECK_IN(y, torch::kFloat32);
+  CHECK_IN(grad, torch::kFloat64);
+  CHECK_IN(rows, torch::kFloat32);
+  CHECK_IN(sol, torch::kFloat64);
+  CHECK_IN(krow, torch::kFloat32);
+  launch_smo_update_dev2(X.data_ptr<float>(), y.data_ptr<float>(),
+                         grad.data_ptr<double>(), rows.data_ptr<float>(),
+                         sol.data_ptr<double>(), krow.data_ptr<float>(),
+                         (float)gamma, X.size(0), cur_stream());
+}
+
 static void smo_update(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
                        torch::Tensor rows, double yidai, double yjdaj,
                        double gamma) {
@@ -344,6 +415,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_hist", &rf_hist, "per-node per-feature class histograms (tree build)");
   m.def("smo_solve", &smo_solve, "device-side SMO pair solve (fused iteration)");
   m.def("smo_update_dev", &smo_update_dev, "gradient update from device sol buffer");
+  m.def("smo_row", &smo_row, "K(x_i, .) kernel row for WSS-2");
+  m.def("smo_select2", &smo_select2, "WSS-2 second-order j selection");
+  m.def("smo_solve2", &smo_solve2, "device-side WSS-2 pair solve");
+  m.def("smo_update_dev2", &smo_update_dev2, "gradient update reusing the WSS-2 kernel row");
   m.def("gnb_predict", &gnb_predict, "fused GaussianNB loglik+argmax");
   m.def("linear_argmax", &linear_argmax, "logits+argmax");
   m.def("kmeans_assign", &kmeans_assign, "Lloyd assign + partial update");

@@ -94,21 +94,43 @@ def _reconstruct_grad(X, y, alpha, grad, gamma, rows_per=1_500_000):
 
 
 def _smo_fused_gpu(X, y, alpha, grad, C, gamma, tol, max_iter, chunk=128,
-                   recompute_every=8192):
+                   recompute_every=8192, wss2=False):
     """Single-GPU fast path: the whole select→solve→update iteration runs on
     device (csrc smo_solve updates alpha in place and re-arms the select
     buffer), so the host only polls the convergence status once per
-    ``chunk`` iterations instead of 5 round-trips per iteration."""
+    ``chunk`` iterations instead of 5 round-trips per iteration.
+
+    ``wss2=True`` switches to libsvm's second-order working-set selection
+    (smo_row / smo_select2 / smo_solve2 / smo_update_dev2): i's kernel row
+    is computed once, drives the second-order j choice, and serves as the
+    i-half of the gradient update.  MEASURED OFF by default: on flow-stat
+    rows with gamma='scale' the RBF kernel values sit near 1 (tiny gamma),
+    the curvature a_t = 2 − 2 y_i y_t K_it degenerates to ~tau for
+    same-class candidates, and WSS-2 cut iterations only ~11% while the two
+    extra n-row passes cost ~35%/iteration — a net loss
+    (profiles/svc_wss2_r02.md).  The path stays built, tested and available
+    for kernels with real curvature spread."""
     from ..ops import gpu as og
 
     device = X.device
-    sel = torch.zeros(2, dtype=torch.int64, device=device)
+    sel = torch.zeros(3, dtype=torch.int64, device=device)
     rows = torch.zeros(24, dtype=torch.float32, device=device)
     sol = torch.zeros(4, dtype=torch.float64, device=device)
+    krow = (
+        torch.zeros(X.shape[0], dtype=torch.float32, device=device)
+        if wss2 else None
+    )
 
     def one_iter():
-        og._ext.smo_select(y, alpha, grad, float(C), sel)
-        og._ext.smo_solve(X, y, alpha, grad, sel, rows, sol, float(C), float(tol), float(gamma))
+        if wss2:
+            og._ext.smo_select(y, alpha, grad, float(C), sel[:2])
+            og._ext.smo_row(X, sel, sol, krow, float(gamma))
+            og._ext.smo_select2(y, alpha, grad, krow, sel, sol, float(C))
+            og._ext.smo_solve2(X, y, alpha, grad, sel, rows, sol, float(C), float(tol), float(gamma))
+            og._ext.smo_update_dev2(X, y, grad, rows, sol, krow, float(gamma))
+            return
+        og._ext.smo_select(y, alpha, grad, float(C), sel[:2])
+        og._ext.smo_solve(X, y, alpha, grad, sel[:2], rows, sol, float(C), float(tol), float(gamma))
         og._ext.smo_update_dev(X, y, grad, rows, sol, float(gamma))
 
     # capture `chunk` iterations in one hipGraph: every kernel argument is a

@@ -11,6 +11,8 @@
 // per-row feature vector, whose index IS runtime (tree node feature ids),
 // lives in LDS instead.
 
+#include <cstdlib>
+
 #include "common.h"
 
 // ---------------------------------------------------------------------------
@@ -223,7 +225,7 @@ DEV float sel12(const Row12& x, unsigned f) {
 // register (10 bits per class, forests up to 1023 trees) — no memory read
 // at all.  Only mixed leaves (feature byte 0xff) touch the probability
 // table.
-template <int C, bool LDS_NODES, bool LDS_PROBS>
+template <int C, bool LDS_NODES, bool LDS_PROBS, bool LDS_FEAT>
 __launch_bounds__(512, 1)
 __global__ void rf_predict_kernel(const float* __restrict__ X,
                                   const uint2* __restrict__ nodes,
@@ -235,6 +237,12 @@ __global__ void rf_predict_kernel(const float* __restrict__ X,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint2* s_nodes = reinterpret_cast<uint2*>(smem);
   float* s_probs = reinterpret_cast<float*>(s_nodes + (LDS_NODES ? n_nodes : 0));
+  // LDS_FEAT: each thread's 12 features live in LDS at pitch 13 (gcd(13,32)
+  // = 1 -> the 32 lanes of a group land on distinct banks) and the per-node
+  // runtime feature pick is ONE ds_read_b32 instead of sel12's ~10-op
+  // register select tree — trading idle LDS-pipe cycles for the VALU issue
+  // slots this kernel is bound on (VALUBusy 99.2%).
+  float* s_feat = s_probs + (LDS_PROBS ? n_leaves * C : 0);
 
   if (LDS_NODES)
     for (int i = threadIdx.x; i < n_nodes; i += blockDim.x) s_nodes[i] = nodes[i];
@@ -242,11 +250,16 @@ __global__ void rf_predict_kernel(const float* __restrict__ X,
     for (int i = threadIdx.x; i < n_leaves * C; i += blockDim.x)
       s_probs[i] = leaf_proba[i];
   if (LDS_NODES || LDS_PROBS) __syncthreads();
+  float* my_feat = s_feat + (int)threadIdx.x * 13;
 
   long long stride = (long long)gridDim.x * blockDim.x;
   for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x; row < n;
        row += stride) {
     Row12 x = load_row12(X, row);
+    if (LDS_FEAT) {
+#pragma unroll
+      for (int j = 0; j < 12; ++j) my_feat[j] = x.v[j];
+    }
     float acc[C];
 #pragma unroll
     for (int c = 0; c < C; ++c) acc[c] = 0.f;
@@ -270,7 +283,8 @@ __global__ void rf_predict_kernel(const float* __restrict__ X,
             break;
           }
           float thr = __uint_as_float(node.x);
-          idx = (sel12(x, feat) <= thr) ? idx + 1 : (int)(node.y >> 8);
+          float fv = LDS_FEAT ? my_feat[feat] : sel12(x, feat);
+          idx = (fv <= thr) ? idx + 1 : (int)(node.y >> 8);
         }
       }
       // EXACT early majority exit: each remaining tree adds at most 1.0 to
@@ -380,22 +394,41 @@ extern "C" void launch_rf_predict(const float* X, const unsigned* nodes,
   // (24 waves) per CU.  Leaf probabilities go to LDS only when everything
   // still fits the 3-block budget; big forests fall back to the
   // (L2-resident) global tables.
+  //
+  // Feature placement (TCSDN_RF_MODE, A/B'd on hardware — 100-tree
+  // reference forest, 10M rows: mode 0 = 2.09 Gflows/s, mode 1 = 2.12,
+  // mode 2 = 2.60): mode 0 keeps features in registers (sel12 select tree,
+  // nodes LDS when they fit); mode 1 also stages features in LDS (pitch 13,
+  // one ds_read per node) but loses a block of occupancy next to a big
+  // forest; mode 2 — THE DEFAULT — stages features in LDS and leaves the
+  // node table in L2 (42 KB forest ≪ 4 MB per-XCD L2; the ~200-cycle node
+  // fetch hides behind 32 waves/CU while the VALU sheds the ~10-op select
+  // tree this kernel was issue-bound on).
+  int mode = 2;
+  if (const char* e = getenv("TCSDN_RF_MODE")) mode = atoi(e);
   size_t node_bytes = (size_t)n_nodes * sizeof(uint2);
   size_t prob_bytes = (size_t)n_leaves * C * sizeof(float);
+  size_t feat_bytes = (size_t)block * 13 * sizeof(float);
   size_t budget = 52 * 1024;  // 3 blocks/CU floor
-  bool lds_nodes = node_bytes <= budget;
-  bool lds_probs = lds_nodes && (node_bytes + prob_bytes) <= budget;
-  size_t lds = (lds_nodes ? node_bytes : 0) + (lds_probs ? prob_bytes : 0);
+  bool lds_feat = mode != 0;
+  bool lds_nodes = mode != 2 && node_bytes + (lds_feat ? feat_bytes : 0) <=
+                                    (mode == 1 ? (size_t)76 * 1024 : budget);
+  bool lds_probs = lds_nodes && !lds_feat && (node_bytes + prob_bytes) <= budget;
+  size_t lds = (lds_nodes ? node_bytes : 0) + (lds_probs ? prob_bytes : 0) +
+               (lds_feat ? feat_bytes : 0);
   dim3 grid(ts_grid(n, block, 4096));
-#define RF_LAUNCH(CV, LN, LP)                                                \
-  hipLaunchKernelGGL((rf_predict_kernel<CV, LN, LP>), grid, dim3(block),     \
+#define RF_LAUNCH(CV, LN, LP, LF)                                            \
+  hipLaunchKernelGGL((rf_predict_kernel<CV, LN, LP, LF>), grid, dim3(block), \
                      lds, stream, X, reinterpret_cast<const uint2*>(nodes),  \
                      roots, leaf_proba, out, n, n_nodes, n_leaves, T)
 #define RF_CASE(CV)                                                          \
   case CV:                                                                   \
-    if (lds_nodes && lds_probs) RF_LAUNCH(CV, true, true);                   \
-    else if (lds_nodes) RF_LAUNCH(CV, true, false);                          \
-    else RF_LAUNCH(CV, false, false);                                        \
+    if (lds_feat) {                                                          \
+      if (lds_nodes) RF_LAUNCH(CV, true, false, true);                       \
+      else RF_LAUNCH(CV, false, false, true);                                \
+    } else if (lds_nodes && lds_probs) RF_LAUNCH(CV, true, true, false);     \
+    else if (lds_nodes) RF_LAUNCH(CV, true, false, false);                   \
+    else RF_LAUNCH(CV, false, false, false);                                 \
     return;
   switch (C) {
     RF_CASE(2) RF_CASE(3) RF_CASE(4) RF_CASE(5) RF_CASE(6) RF_CASE(7)

@@ -18,6 +18,8 @@ REQUIRED = [
 CASES = {
     "rf": ["--steps", "2", "--warmup", "1"],
     "knn": ["--workload", "knn", "--steps", "2", "--warmup", "1", "--knn-queries", "2048"],
+    "knn-approx": ["--workload", "knn", "--knn-approx", "--steps", "2", "--warmup", "1",
+                   "--knn-queries", "2048"],
     "svc-fit": ["--workload", "svc-fit", "--steps", "2", "--warmup", "1", "--svc-iters-per-step", "5"],
     "svc-fit-full": ["--workload", "svc-fit-full", "--steps", "1", "--warmup", "0",
                      "--svc-rows", "1500", "--svc-full-max-iter", "60"],

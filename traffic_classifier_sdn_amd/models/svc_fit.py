@@ -62,7 +62,7 @@ def _grad_update_cpu(X, y, grad, xi, xj, yidai, yjdaj, gamma):
     grad += y.double() * (yidai * ki + yjdaj * kj)
 
 
-def _reconstruct_grad(X, y, alpha, grad, gamma, rows_per=1_500_000):
+def _reconstruct_grad(X, y, alpha, grad, gamma):
     """Rebuild grad_i = y_i·Σ_j α_j y_j K(i,j) − 1 from scratch (f64).
 
     The fused iteration updates the gradient incrementally with two f32

@@ -216,3 +216,25 @@ def test_fit_with_synth_quake_six_classes(tmp_path):
 
     m = load_model(str(tmp_path / "GaussianNB.npz"))
     assert sorted(str(c) for c in m.classes_) == ["dns", "game", "ping", "quake", "telnet", "voice"]
+
+
+def test_fit_torchrun_world2_gloo(tmp_path):
+    """fit.py end-to-end under torchrun (2 gloo ranks): sharded DP fit, rank
+    0 writes the checkpoint — the launch mode an 8-GPU node uses."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29641", "-m", "traffic_classifier_sdn_amd.fit",
+         "--algos", "gaussiannb", "--out", str(tmp_path), "--json"],
+        capture_output=True, text=True, timeout=420, cwd=repo,
+        env={**os.environ, "OMP_NUM_THREADS": "1"},
+    )
+    assert r.returncode == 0, r.stderr[-1200:]
+    assert (tmp_path / "GaussianNB.npz").exists()
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["accuracy"] > 0.97

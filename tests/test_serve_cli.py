@@ -227,3 +227,30 @@ def test_cli_subprocess_source_with_custom_monitor(tmp_path):
     assert r.returncode == 0, r.stderr[-1500:]
     assert "Traffic Type" in r.stdout
     assert r.stdout.count("Flow ID") >= 2  # multiple prediction passes
+
+
+def test_prometheus_metrics_endpoint():
+    """--prometheus exposes flow/latency metrics on a scrape port."""
+    import urllib.request
+
+    pytest.importorskip("prometheus_client")
+    from traffic_classifier_sdn_amd.flow.replay import TelemetryReplaySource
+    from traffic_classifier_sdn_amd.models import GaussianNB
+    from traffic_classifier_sdn_amd.serve import RealtimeClassifier
+    from traffic_classifier_sdn_amd.utils.datasets import load_reference_dataset
+
+    X, y = load_reference_dataset()
+    m = GaussianNB().fit(X[:500], y[:500])
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    rc = RealtimeClassifier(m, out=io.StringIO(), prometheus_port=port)
+    for line in TelemetryReplaySource(seed=1).stream(6):
+        rc.feed(line)
+    body = urllib.request.urlopen(f"http://127.0.0.1:{port}/metrics", timeout=10).read().decode()
+    assert "tcsdn_flows_tracked" in body
+    assert "tcsdn_predict_passes_total" in body
+    assert "tcsdn_predict_seconds_bucket" in body
+    assert "tcsdn_class_flows" in body

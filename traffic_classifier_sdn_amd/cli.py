@@ -94,6 +94,8 @@ def main(argv: Optional[list] = None) -> int:
     parser.add_argument("--timeout", type=int, default=DEFAULT_TIMEOUT, help="train collection seconds")
     parser.add_argument("--replay-polls", type=int, default=30)
     parser.add_argument("--seed", type=int, default=0)
+    parser.add_argument("--prometheus", type=int, default=0, metavar="PORT",
+                        help="expose serve metrics on a Prometheus scrape port")
     parser.add_argument("--stats", action="store_true",
                         help="emit one JSON perf line (flows, predict_ms, flows/sec) per prediction pass on stderr")
     args = parser.parse_args(argv)
@@ -132,7 +134,7 @@ def main(argv: Optional[list] = None) -> int:
         print(f"ERROR: checkpoint {path} not found", file=sys.stderr)
         return 2
     model = load_model(path, device=args.device)
-    rc = RealtimeClassifier(model, stats=args.stats)
+    rc = RealtimeClassifier(model, stats=args.stats, prometheus_port=args.prometheus)
     try:
         rc.run(_line_source(args))
     except KeyboardInterrupt:

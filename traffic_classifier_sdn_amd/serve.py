@@ -59,6 +59,7 @@ class RealtimeClassifier:
         out: TextIO = sys.stdout,
         stats: bool = False,
         stats_out: TextIO = sys.stderr,
+        prometheus_port: int = 0,
     ) -> None:
         self.model = model
         self.parser = PollStreamParser()
@@ -67,6 +68,11 @@ class RealtimeClassifier:
         self.stats = stats
         self.stats_out = stats_out
         self._last_batch = 0
+        self.prom = None
+        if prometheus_port:
+            from .utils.prom import PromServeMetrics
+
+            self.prom = PromServeMetrics(prometheus_port)
 
     def classify_now(self) -> np.ndarray:
         table = self.parser.table
@@ -103,6 +109,10 @@ class RealtimeClassifier:
             predict_s = time.perf_counter() - t0
             self.out.write(render_flow_table(self.parser.table, labels) + "\n")
             self.out.flush()
+            if self.prom is not None:
+                self.prom.observe_pass(
+                    len(self.parser.table), self.parser.records, predict_s, labels
+                )
             if self.stats:
                 n = len(self.parser.table)
                 self.stats_out.write(

@@ -316,6 +316,133 @@ __global__ void rf_predict_kernel(const float* __restrict__ X,
   }
 }
 
+// ILP-2 variant (mode 3): two independent traversal chains per lane.  At
+// mode 2 the kernel is latency-bound (VALUBusy 57.5%, SQ_WAIT_ANY 68% of
+// wave cycles parked on the L2 node fetch + LDS feature read chains) and
+// the 32-waves/CU cap is already reached — more memory-level parallelism
+// has to come from within the wave.  256-thread blocks keep the two
+// pitch-13 feature slabs at the same 26.6 KB the 512-thread single-row
+// variant used, so occupancy is unchanged and each CU runs 2x the chains.
+template <int C>
+__launch_bounds__(256, 1)
+__global__ void rf_predict_ilp2_kernel(const float* __restrict__ X,
+                                       const uint2* __restrict__ nodes,
+                                       const int* __restrict__ roots,
+                                       const float* __restrict__ leaf_proba,
+                                       int* __restrict__ out, long long n,
+                                       int T) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* s_feat = reinterpret_cast<float*>(smem);
+  float* fA = s_feat + (int)threadIdx.x * 13;
+  float* fB = s_feat + (int)(blockDim.x + threadIdx.x) * 13;
+
+  long long stride = (long long)gridDim.x * blockDim.x * 2;
+  for (long long base = (long long)blockIdx.x * blockDim.x * 2; base < n;
+       base += stride) {
+    const long long rowA = base + threadIdx.x;
+    const long long rowB = base + blockDim.x + threadIdx.x;
+    if (rowA < n) {
+      Row12 x = load_row12(X, rowA);
+#pragma unroll
+      for (int j = 0; j < 12; ++j) fA[j] = x.v[j];
+    }
+    if (rowB < n) {
+      Row12 x = load_row12(X, rowB);
+#pragma unroll
+      for (int j = 0; j < 12; ++j) fB[j] = x.v[j];
+    }
+    float accA[C], accB[C];
+#pragma unroll
+    for (int c = 0; c < C; ++c) accA[c] = accB[c] = 0.f;
+    unsigned long long votesA = 0ull, votesB = 0ull;
+    int decidedA = rowA >= n, decidedB = rowB >= n;
+    for (int t = 0; t < T; ++t) {
+      int idxA = roots[t], idxB = idxA;
+      bool runA = !decidedA, runB = !decidedB;
+      while (runA || runB) {
+        uint2 nodeA, nodeB;
+        if (runA) nodeA = nodes[idxA];
+        if (runB) nodeB = nodes[idxB];
+        if (runA) {
+          unsigned feat = nodeA.y & 0xffu;
+          if (feat >= 0xf0u) {
+            if (feat == 0xffu) {
+              int pr = (int)nodeA.x * C;
+#pragma unroll
+              for (int c = 0; c < C; ++c) accA[c] += leaf_proba[pr + c];
+            } else {
+              votesA += 1ull << ((feat & 0xfu) * 10);
+            }
+            runA = false;
+          } else {
+            float thr = __uint_as_float(nodeA.x);
+            idxA = (fA[feat] <= thr) ? idxA + 1 : (int)(nodeA.y >> 8);
+          }
+        }
+        if (runB) {
+          unsigned feat = nodeB.y & 0xffu;
+          if (feat >= 0xf0u) {
+            if (feat == 0xffu) {
+              int pr = (int)nodeB.x * C;
+#pragma unroll
+              for (int c = 0; c < C; ++c) accB[c] += leaf_proba[pr + c];
+            } else {
+              votesB += 1ull << ((feat & 0xfu) * 10);
+            }
+            runB = false;
+          } else {
+            float thr = __uint_as_float(nodeB.x);
+            idxB = (fB[feat] <= thr) ? idxB + 1 : (int)(nodeB.y >> 8);
+          }
+        }
+      }
+      if ((t & 7) == 7) {
+        if (!decidedA) {
+          float m1 = -INFINITY, m2 = -INFINITY;
+#pragma unroll
+          for (int c = 0; c < C; ++c) {
+            float sc = accA[c] + (float)((votesA >> (c * 10)) & 1023ull);
+            if (sc > m1) { m2 = m1; m1 = sc; }
+            else if (sc > m2) m2 = sc;
+          }
+          if (m1 - m2 > (float)(T - 1 - t)) decidedA = 1;
+        }
+        if (!decidedB) {
+          float m1 = -INFINITY, m2 = -INFINITY;
+#pragma unroll
+          for (int c = 0; c < C; ++c) {
+            float sc = accB[c] + (float)((votesB >> (c * 10)) & 1023ull);
+            if (sc > m1) { m2 = m1; m1 = sc; }
+            else if (sc > m2) m2 = sc;
+          }
+          if (m1 - m2 > (float)(T - 1 - t)) decidedB = 1;
+        }
+        if (__all(decidedA && decidedB)) break;
+      }
+    }
+    if (rowA < n) {
+      float best = -INFINITY;
+      int bi = 0;
+#pragma unroll
+      for (int c = 0; c < C; ++c) {
+        float sc = accA[c] + (float)((votesA >> (c * 10)) & 1023ull);
+        if (sc > best) { best = sc; bi = c; }
+      }
+      out[rowA] = bi;
+    }
+    if (rowB < n) {
+      float best = -INFINITY;
+      int bi = 0;
+#pragma unroll
+      for (int c = 0; c < C; ++c) {
+        float sc = accB[c] + (float)((votesB >> (c * 10)) & 1023ull);
+        if (sc > best) { best = sc; bi = c; }
+      }
+      out[rowB] = bi;
+    }
+  }
+}
+
 // Small-batch variant: ONE WAVE PER ROW, trees split across lanes.  At
 // serve batch sizes the row-per-lane kernel runs a ~900-step serial
 // dependent-load chain per row on a mostly idle chip; here each lane walks
@@ -406,6 +533,23 @@ extern "C" void launch_rf_predict(const float* X, const unsigned* nodes,
   // tree this kernel was issue-bound on).
   int mode = 2;
   if (const char* e = getenv("TCSDN_RF_MODE")) mode = atoi(e);
+  if (mode == 3) {  // ILP-2: two chains per lane, 256-thread blocks
+    const int b2 = 256;
+    size_t lds2 = (size_t)b2 * 2 * 13 * sizeof(float);
+    dim3 g2(ts_grid((n + 1) / 2, b2, 4096));
+#define RFI_CASE(CV)                                                         \
+  case CV:                                                                   \
+    hipLaunchKernelGGL((rf_predict_ilp2_kernel<CV>), g2, dim3(b2), lds2,     \
+                       stream, X, reinterpret_cast<const uint2*>(nodes),     \
+                       roots, leaf_proba, out, n, T);                        \
+    return;
+    switch (C) {
+      RFI_CASE(2) RFI_CASE(3) RFI_CASE(4) RFI_CASE(5) RFI_CASE(6) RFI_CASE(7)
+      RFI_CASE(8) RFI_CASE(12) RFI_CASE(16)
+      default: break;
+    }
+#undef RFI_CASE
+  }
   size_t node_bytes = (size_t)n_nodes * sizeof(uint2);
   size_t prob_bytes = (size_t)n_leaves * C * sizeof(float);
   size_t feat_bytes = (size_t)block * 13 * sizeof(float);

@@ -8,15 +8,21 @@ Rank 0 prints ONE JSON line; elapsed is the MAX over ranks, timed region
 bracketed by barrier + torch.cuda.synchronize on both sides.
 
 Extra workloads (same JSON contract, run explicitly):
-    --workload knn       config #4: k=5 brute-force, reference set sharded
-                         across ranks (12.5M rows/GPU -> 100M at 8), per-step
-                         top-k + RCCL all-gather merge over 64K query rows
-    --workload svc-fit   config #3: RBF-SVC SMO on 1M rows strong-scaled
-                         across ranks; per step a fixed block of SMO
-                         iterations (fused kernel-row gradient updates)
-    --workload serve     config #5: full poll cycle — counter snapshot ->
-                         H2D -> hipGraph replay of feature-extract + 5-model
-                         ensemble predict -> D2H labels, 8192 live flows
+    --workload knn           config #4: k=5 brute-force, reference set
+                             sharded across ranks (12.5M rows/GPU -> 100M at
+                             8), per-step top-k + RCCL all-gather merge over
+                             64K query rows; --knn-approx switches to the
+                             bf16 coarse-pass selection (exact refine,
+                             measured recall)
+    --workload svc-fit       config #3 inner loop: fixed block of fused SMO
+                             iterations on 1M rows, strong-scaled
+    --workload svc-fit-full  config #3 proper: the full 6-class one-vs-one
+                             fit, all 15 pairs to tolerance, held-out
+                             accuracy reported untimed
+    --workload serve         config #5: full poll cycle — counter snapshot
+                             -> H2D -> hipGraph replay of feature-extract +
+                             5-model ensemble predict -> D2H labels
+    --workload rf-fit        level-synchronous histogram forest build
 
 The forest/model shapes are the reference's own checkpoints (converted to
 data/ref_models/*.npz), so model config matches the named baseline; data is

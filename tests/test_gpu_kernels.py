@@ -369,7 +369,7 @@ def test_svc_predict_large_nsv_accumulation():
     from traffic_classifier_sdn_amd.ops.gpu import _ext
 
     rng = np.random.default_rng(11)
-    C, nsv_per, n = 6, 6000, 40_000  # n > 32768 -> tiled kernel
+    C, nsv_per, n = 6, 6000, 150_000  # n > 131072 -> tiled kernel
     nsv = C * nsv_per
     SV = torch.from_numpy(rng.normal(size=(nsv, 12))).float()
     # bounded-alpha-style duals: mostly +-1 (the pathological regime)

@@ -501,7 +501,7 @@ extern "C" void launch_rf_predict(const float* X, const unsigned* nodes,
                                   int* out, long long n, int n_nodes,
                                   int n_leaves, int T, int C,
                                   hipStream_t stream) {
-  if (n <= 32768) {  // wave-per-row fills the chip at serve batch sizes
+  if (n <= 131072) {  // wave-per-row fills the chip at serve batch sizes
     dim3 wgrid((unsigned)((n + 3) / 4));
 #define RFW_CASE(CV)                                                        \
   case CV:                                                                  \
@@ -766,7 +766,10 @@ extern "C" void launch_svc_predict(const float* X, const float* SV,
                                    long long n, int nsv, int C, float gamma,
                                    hipStream_t stream) {
   const int block = 256;
-  if (n <= 32768) {  // wave-per-row fills the chip at serve batch sizes
+  // crossover re-measured in round 2 after the f64-accumulator fix: at 65K
+  // rows the wave-per-row kernel beats the tiled one (serve 34.5 -> 39.6M
+  // flows/s), so the cutover moved 32768 -> 131072 (same for RF above)
+  if (n <= 131072) {  // wave-per-row fills the chip at serve batch sizes
     dim3 wgrid((unsigned)((n + 3) / 4));
 #define SVC_WCASE(CV)                                                       \
   case CV:                                                                  \

@@ -455,3 +455,26 @@ def test_smo_wss2_converges():
     d1 = K @ (a1 * yv.double()) + b1
     d2 = K @ (a2 * yv.double()) + b2
     assert float(((d1 > 0) == (d2 > 0)).float().mean()) > 0.995
+
+
+@pytest.mark.gpu
+def test_predict_parity_wave_tiled_band(X_real):
+    """The RF/SVC wave-per-row kernels now serve up to 131072 rows (cutover
+    re-tuned in round 2); pin prediction parity in the 32K-131K band the
+    old threshold never exercised, and across the boundary."""
+    from traffic_classifier_sdn_amd.models import load_model
+    from traffic_classifier_sdn_amd.utils.datasets import synthetic_flow_rows
+
+    for n in (65_536, 140_000):  # wave side / tiled side of the cutover
+        X = torch.from_numpy(synthetic_flow_rows(n, seed=21)).float()
+        rf = load_model(os.path.join(REPO, "data", "ref_models", "RandomForestClassifier.npz"), device="cuda")
+        got = rf.predict_index(X.cuda()).cpu()
+        rf_cpu = load_model(os.path.join(REPO, "data", "ref_models", "RandomForestClassifier.npz"))
+        want = rf_cpu.predict_index(X)
+        assert (got == want).float().mean().item() > 0.9995, n
+
+        svc = load_model(os.path.join(REPO, "data", "ref_models", "SVC.npz"), device="cuda")
+        got_s = svc.predict_index(X.cuda()).cpu()
+        svc_cpu = load_model(os.path.join(REPO, "data", "ref_models", "SVC.npz"))
+        want_s = svc_cpu.predict_index(X)
+        assert (got_s == want_s).float().mean().item() > 0.999, n

@@ -469,12 +469,12 @@ def test_predict_parity_wave_tiled_band(X_real):
         X = torch.from_numpy(synthetic_flow_rows(n, seed=21)).float()
         rf = load_model(os.path.join(REPO, "data", "ref_models", "RandomForestClassifier.npz"), device="cuda")
         got = rf.predict_index(X.cuda()).cpu()
-        rf_cpu = load_model(os.path.join(REPO, "data", "ref_models", "RandomForestClassifier.npz"))
+        rf_cpu = load_model(os.path.join(REPO, "data", "ref_models", "RandomForestClassifier.npz"), device="cpu")
         want = rf_cpu.predict_index(X)
         assert (got == want).float().mean().item() > 0.9995, n
 
         svc = load_model(os.path.join(REPO, "data", "ref_models", "SVC.npz"), device="cuda")
         got_s = svc.predict_index(X.cuda()).cpu()
-        svc_cpu = load_model(os.path.join(REPO, "data", "ref_models", "SVC.npz"))
+        svc_cpu = load_model(os.path.join(REPO, "data", "ref_models", "SVC.npz"), device="cpu")
         want_s = svc_cpu.predict_index(X)
         assert (got_s == want_s).float().mean().item() > 0.999, n

@@ -33,7 +33,7 @@ import torch
 from .. import ops
 
 N_BINS = 256
-_CHUNK_NODES = 2048  # per rf_hist pass: 2048 * 73.7 KB ~ 150 MB of histogram
+_CHUNK_NODES = 8192  # per rf_hist pass: 8192 * 73.7 KB ~ 600 MB of histogram (each extra chunk re-scans ALL rows, so size the chunk to hold a whole combined multi-tree frontier)
 
 
 def quantize(X: torch.Tensor, max_sample: int = 262_144, seed: int = 0):
@@ -52,40 +52,8 @@ def quantize(X: torch.Tensor, max_sample: int = 262_144, seed: int = 0):
     return bins, edges
 
 
-def build_forest_hist(
-    X: torch.Tensor,
-    y_idx: torch.Tensor,
-    n_classes: int,
-    n_estimators: int,
-    max_features: int = 3,
-    seed: Optional[int] = 0,
-    bootstrap: bool = True,
-    max_depth: int = 40,
-    min_samples_split: int = 2,
-    tree_range=None,
-) -> List[Dict[str, np.ndarray]]:
-    device = X.device
-    n, F = X.shape
-    bins, edges = quantize(X, seed=0 if seed is None else seed)
-    y8 = y_idx.to(torch.uint8).contiguous().to(device)
-    edges_np = edges.cpu().numpy()
-    trees = []
-    for t in tree_range if tree_range is not None else range(n_estimators):
-        g = torch.Generator(device="cpu").manual_seed((0 if seed is None else seed) + 1000003 * t)
-        if bootstrap:
-            rows = torch.randint(0, n, (n,), generator=g).to(device)
-            B = bins[rows].contiguous()
-            Y = y8[rows].contiguous()
-        else:
-            B, Y = bins, y8
-        trees.append(
-            _build_one(B, Y, n_classes, edges_np, g, max_features, max_depth, min_samples_split)
-        )
-    return trees
-
-
 def _level_split_search(B, Y, nid, L, C, mf, g, device):
-    """Chunked histogram + split search over the frontier.
+    """Chunked histogram + split search over the (combined) frontier.
 
     Returns numpy arrays over the L frontier nodes:
     (counts [L,C], best_imp [L], best_f [L], best_b [L]).
@@ -135,31 +103,90 @@ def _level_split_search(B, Y, nid, L, C, mf, g, device):
     return cnt_all, imp_all, f_all, b_all
 
 
-def _build_one(B, Y, C, edges_np, g, mf, max_depth, min_split) -> Dict[str, np.ndarray]:
-    device = B.device
-    n, F = B.shape
-    nid = torch.zeros(n, dtype=torch.int32, device=device)  # frontier-LOCAL ids
+def build_forest_hist(
+    X: torch.Tensor,
+    y_idx: torch.Tensor,
+    n_classes: int,
+    n_estimators: int,
+    max_features: int = 3,
+    seed: Optional[int] = 0,
+    bootstrap: bool = True,
+    max_depth: int = 40,
+    min_samples_split: int = 2,
+    tree_range=None,
+) -> List[Dict[str, np.ndarray]]:
+    """Grow ALL trees of the (local) forest level-synchronously.
 
-    # per-level records (vectorised; concatenated at the end)
-    lvl_ids: List[np.ndarray] = []      # global ids of the level's nodes
-    lvl_vals: List[np.ndarray] = []     # class counts [L, C]
-    lvl_feat: List[np.ndarray] = []     # split feature (-1 leaf)
+    Round-2 rewrite: the first version built trees one at a time — 25 trees
+    x ~12 levels = ~300 sequential {hist kernel + split search + partition}
+    iterations, so the fit was bound by per-level Python/launch/D2H glue,
+    not the scatter kernel (9% of step time, profiles/pmc_counters_r02.md).
+    Now every tree's frontier lives in ONE combined node numbering: each
+    level runs ONE rf_hist scatter over the T*n bootstrap rows, one
+    vectorised split search, and one partition pass — ~12 iterations total
+    for any tree count.  The combined frontier stays sorted by tree (stable
+    child emission), so per-tree records fall out by masking at the end.
+    """
+    device = X.device
+    n, F = X.shape
+    seed0 = 0 if seed is None else seed
+    bins, edges = quantize(X, seed=seed0)
+    y8 = y_idx.to(torch.uint8).contiguous().to(device)
+    edges_np = edges.cpu().numpy()
+    tree_list = list(tree_range) if tree_range is not None else list(range(n_estimators))
+    T = len(tree_list)
+    if T == 0:
+        return []
+
+    # per-tree bootstrap resample (same per-tree generators/seeds as the
+    # sequential builder so resamples are reproducible per tree id)
+    if bootstrap:
+        rows = torch.cat(
+            [
+                torch.randint(
+                    0, n, (n,),
+                    generator=torch.Generator(device="cpu").manual_seed(seed0 + 1000003 * t),
+                )
+                for t in tree_list
+            ]
+        ).to(device)
+        Ball = bins[rows].contiguous()
+        Yall = y8[rows].contiguous()
+    else:
+        Ball = bins.repeat(T, 1).contiguous()
+        Yall = y8.repeat(T).contiguous()
+
+    gsel = torch.Generator(device="cpu").manual_seed(seed0 * 7919 + 13)
+    # combined-frontier-LOCAL node id per row; tree t's root is slot t
+    nid = torch.repeat_interleave(
+        torch.arange(T, dtype=torch.int32), n
+    ).to(device)
+
+    f_tree = np.arange(T, dtype=np.int64)       # tree of each frontier node
+    f_gid = np.zeros(T, dtype=np.int64)         # per-tree global node id
+    tot = np.ones(T, dtype=np.int64)            # per-tree node counts
+
+    # per-level combined records (split per tree at the end)
+    lvl_tree: List[np.ndarray] = []
+    lvl_ids: List[np.ndarray] = []
+    lvl_vals: List[np.ndarray] = []
+    lvl_feat: List[np.ndarray] = []
     lvl_thr: List[np.ndarray] = []
-    lvl_left: List[np.ndarray] = []     # global child ids (-1 leaf)
+    lvl_left: List[np.ndarray] = []
     lvl_right: List[np.ndarray] = []
-    frontier_ids = np.array([0], dtype=np.int64)
-    tot = 1
 
     for depth in range(max_depth):
-        L = len(frontier_ids)
+        L = len(f_tree)
         if L == 0:
             break
-        cnt, best_imp, best_f, best_b = _level_split_search(B, Y, nid, L, C, mf, g, device)
+        cnt, best_imp, best_f, best_b = _level_split_search(
+            Ball, Yall, nid, L, n_classes, max_features, gsel, device
+        )
         n_node = cnt.sum(axis=1)
         pp = cnt / np.maximum(n_node, 1.0)[:, None]
         parent_gini = 1.0 - (pp * pp).sum(axis=1)
         can = (
-            (n_node >= min_split)
+            (n_node >= min_samples_split)
             & np.isfinite(best_imp)
             & (best_imp < parent_gini - 1e-12)
             & (depth < max_depth - 1)
@@ -167,21 +194,28 @@ def _build_one(B, Y, C, edges_np, g, mf, max_depth, min_split) -> Dict[str, np.n
         n_split = int(can.sum())
         feat = np.where(can, best_f, -1).astype(np.int64)
         thr = np.where(can, edges_np[np.minimum(best_f, F - 1), best_b], 0.0)
+        # per-tree child global ids: frontier is tree-sorted, so the
+        # within-tree split rank is positional
+        ctree = f_tree[can]
+        counts_t = np.bincount(ctree, minlength=T)
+        starts_t = np.concatenate([[0], np.cumsum(counts_t)[:-1]])
+        rank_t = np.arange(n_split, dtype=np.int64) - starts_t[ctree]
+        kids = tot[ctree] + 2 * rank_t
         left = np.full(L, -1, dtype=np.int64)
         right = np.full(L, -1, dtype=np.int64)
-        kids = tot + 2 * np.arange(n_split, dtype=np.int64)
         left[can] = kids
         right[can] = kids + 1
-        lvl_ids.append(frontier_ids)
+        lvl_tree.append(f_tree)
+        lvl_ids.append(f_gid)
         lvl_vals.append(cnt)
         lvl_feat.append(feat)
         lvl_thr.append(thr)
         lvl_left.append(left)
         lvl_right.append(right)
         if n_split == 0:
-            frontier_ids = np.array([], dtype=np.int64)
+            f_tree = np.array([], dtype=np.int64)
             break
-        # row partition: local id -> new local id of the LEFT child
+        # row partition: combined local id -> new local id of the LEFT child
         lmap = np.full(L, -1, dtype=np.int64)
         lmap[can] = 2 * np.arange(n_split, dtype=np.int64)
         lmap_t = torch.from_numpy(lmap).to(device)
@@ -190,18 +224,43 @@ def _build_one(B, Y, C, edges_np, g, mf, max_depth, min_split) -> Dict[str, np.n
         live = nid >= 0
         nid_l = nid.long().clamp(min=0)
         splitting = live & (lmap_t[nid_l] >= 0)
-        vals = B.gather(1, feat_t[nid_l].unsqueeze(1)).squeeze(1).long()
+        vals = Ball.gather(1, feat_t[nid_l].unsqueeze(1)).squeeze(1).long()
         go_left = vals <= bin_t[nid_l]
         child = lmap_t[nid_l] + torch.where(go_left, 0, 1)
         nid = torch.where(splitting, child.to(torch.int32), torch.full_like(nid, -1))
-        # next level's global ids, in local order [l0, r0, l1, r1, ...]
-        nxt = np.empty(2 * n_split, dtype=np.int64)
-        nxt[0::2] = kids
-        nxt[1::2] = kids + 1
-        frontier_ids = nxt
-        tot += 2 * n_split
+        # next frontier, combined-local order [l0, r0, l1, r1, ...]
+        nxt_gid = np.empty(2 * n_split, dtype=np.int64)
+        nxt_gid[0::2] = kids
+        nxt_gid[1::2] = kids + 1
+        nxt_tree = np.empty(2 * n_split, dtype=np.int64)
+        nxt_tree[0::2] = ctree
+        nxt_tree[1::2] = ctree
+        f_gid = nxt_gid
+        f_tree = nxt_tree
+        tot += 2 * counts_t
 
-    # flatten per-level records into id-indexed arrays
+    # split the combined records per tree and flatten each
+    trees: List[Dict[str, np.ndarray]] = []
+    for t in range(T):
+        ids_l, vals_l, feat_l, thr_l, left_l, right_l = [], [], [], [], [], []
+        for d in range(len(lvl_ids)):
+            m = lvl_tree[d] == t
+            if not m.any():
+                continue
+            ids_l.append(lvl_ids[d][m])
+            vals_l.append(lvl_vals[d][m])
+            feat_l.append(lvl_feat[d][m])
+            thr_l.append(lvl_thr[d][m])
+            left_l.append(lvl_left[d][m])
+            right_l.append(lvl_right[d][m])
+        trees.append(
+            _flatten_tree(int(tot[t]), n_classes, ids_l, vals_l, feat_l, thr_l, left_l, right_l)
+        )
+    return trees
+
+
+def _flatten_tree(tot, C, lvl_ids, lvl_vals, lvl_feat, lvl_thr, lvl_left, lvl_right) -> Dict[str, np.ndarray]:
+    """Per-level records -> id-indexed arrays -> depth-first preorder."""
     feature = np.full(tot, -1, dtype=np.int64)
     thr = np.zeros(tot, dtype=np.float64)
     left = np.full(tot, -1, dtype=np.int64)
@@ -214,9 +273,8 @@ def _build_one(B, Y, C, edges_np, g, mf, max_depth, min_split) -> Dict[str, np.n
         right[ids] = r
         values[ids] = v
 
-    # depth-first renumber without a python walk: preorder position =
-    # pos(left) = pos(parent)+1, pos(right) = pos(parent)+1+size(left),
-    # with subtree sizes computed bottom-up per level (vectorised).
+    # preorder renumber: pos(left) = pos(parent)+1,
+    # pos(right) = pos(parent)+1+size(left); sizes bottom-up per level
     size = np.ones(tot, dtype=np.int64)
     for ids, l, r in zip(reversed(lvl_ids), reversed(lvl_left), reversed(lvl_right)):
         has = l >= 0

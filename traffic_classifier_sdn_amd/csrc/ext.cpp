@@ -57,6 +57,8 @@ void launch_smo_update_dev(const float*, const float*, double*, const float*,
                            const double*, float, long long, hipStream_t);
 void launch_smo_row(const float*, const unsigned long long*, const double*,
                     float*, float, long long, hipStream_t);
+void launch_rf_split(const int*, const unsigned char*, unsigned long long*,
+                     int*, int, int, hipStream_t);
 void launch_smo_select2(const float*, const double*, const double*,
                         const float*, unsigned long long*, const double*,
                         double, long long, hipStream_t);
@@ -337,6 +339,22 @@ static void smo_update_dev(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
                         cur_stream());
 }
 
+static std::vector<torch::Tensor> rf_split(torch::Tensor hist,
+                                           torch::Tensor fsel) {
+  CHECK_IN(hist, torch::kInt32);
+  CHECK_IN(fsel, torch::kUInt8);
+  TORCH_CHECK(hist.dim() == 4 && hist.size(1) == 12 && hist.size(2) == 256,
+              "hist must be (L,12,256,C)");
+  int L = (int)hist.size(0);
+  int C = (int)hist.size(3);
+  auto best = torch::full({L}, -1, hist.options().dtype(torch::kInt64));
+  auto cnt = torch::zeros({L, C}, hist.options());
+  launch_rf_split(hist.data_ptr<int>(), fsel.data_ptr<unsigned char>(),
+                  reinterpret_cast<unsigned long long*>(best.data_ptr<int64_t>()),
+                  cnt.data_ptr<int>(), L, C, cur_stream());
+  return {best, cnt};
+}
+
 static void smo_row(torch::Tensor X, torch::Tensor sel, torch::Tensor sol,
                     torch::Tensor krow, double gamma) {
   CHECK_IN(X, torch::kFloat32);
@@ -415,6 +433,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_hist", &rf_hist, "per-node per-feature class histograms (tree build)");
   m.def("smo_solve", &smo_solve, "device-side SMO pair solve (fused iteration)");
   m.def("smo_update_dev", &smo_update_dev, "gradient update from device sol buffer");
+  m.def("rf_split", &rf_split, "fused gini split search over a level histogram");
   m.def("smo_row", &smo_row, "K(x_i, .) kernel row for WSS-2");
   m.def("smo_select2", &smo_select2, "WSS-2 second-order j selection");
   m.def("smo_solve2", &smo_solve2, "device-side WSS-2 pair solve");

@@ -3,6 +3,8 @@
 // host-side L-BFGS / closed-form updates match the f64 CPU oracles, and the
 // resulting buffers are exactly what gets RCCL-all-reduced per step.
 
+#include <cfloat>
+
 #include "common.h"
 
 // ---------------------------------------------------------------------------
@@ -282,6 +284,102 @@ extern "C" void launch_smo_select(const float* y, const double* alpha,
   const int block = 256;
   hipLaunchKernelGGL(smo_select_kernel, dim3(ts_grid(n, block)), dim3(block), 0,
                      stream, y, alpha, grad, C, n, out);
+}
+
+// ---------------------------------------------------------------------------
+// RF split search (round 2): fused gini scan over the level histogram.
+//
+// The torch formulation of the split search materialised a cumsum over the
+// whole [L,12,256,C] histogram plus ~8 further GB-scale intermediates per
+// level; this kernel assigns one THREAD per (node, feature), keeps the
+// running class counts in registers, and resolves the per-node best
+// (impurity, feature, bin) with ONE packed u64 atomicMin per thread —
+// the histogram is read exactly twice (class totals, then the scan).
+//   hist:   int32 [L, F=12, 256, C]
+//   fsel:   uint8 [L, 12]   1 = feature is an mtry candidate for the node
+//   best:   u64   [L]       init ~0; (enc_f32(imp) << 32) | (f << 16) | bin
+//   cnt:    int32 [L, C]    per-node class counts (written by feature 0)
+// Ties break toward the smaller (f, bin) — the same order torch's flat
+// argmin used, so tree construction stays deterministic.
+// ---------------------------------------------------------------------------
+template <int C>
+__global__ void rf_split_kernel(const int* __restrict__ hist,
+                                const unsigned char* __restrict__ fsel,
+                                unsigned long long* __restrict__ best,
+                                int* __restrict__ cnt, int L) {
+  constexpr int F = 12;
+  long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long long)L * F) return;
+  int node = (int)(idx / F);
+  int f = (int)(idx % F);
+  const int* h = hist + ((long long)node * F + f) * 256 * C;
+  float total[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) total[c] = 0.f;
+  for (int b = 0; b < 256; ++b)
+#pragma unroll
+    for (int c = 0; c < C; ++c) total[c] += (float)h[b * C + c];
+  float n_node = 0.f;
+#pragma unroll
+  for (int c = 0; c < C; ++c) n_node += total[c];
+  if (f == 0) {
+#pragma unroll
+    for (int c = 0; c < C; ++c) cnt[node * C + c] = (int)total[c];
+  }
+  if (!fsel[node * F + f]) return;
+  float cum[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) cum[c] = 0.f;
+  float best_imp = FLT_MAX;
+  int best_b = -1;
+  const float inv_n = 1.f / fmaxf(n_node, 1.f);
+  for (int b = 0; b < 255; ++b) {  // bin 255: empty right side, never valid
+    float nl = 0.f;
+#pragma unroll
+    for (int c = 0; c < C; ++c) {
+      cum[c] += (float)h[b * C + c];
+      nl += cum[c];
+    }
+    float nr = n_node - nl;
+    if (nl < 1.f || nr < 1.f) continue;
+    float sl = 0.f, sr = 0.f;
+#pragma unroll
+    for (int c = 0; c < C; ++c) {
+      float pl = cum[c] / nl;
+      float pr = (total[c] - cum[c]) / nr;
+      sl = fmaf(pl, pl, sl);
+      sr = fmaf(pr, pr, sr);
+    }
+    float imp = (nl * (1.f - sl) + nr * (1.f - sr)) * inv_n;
+    if (imp < best_imp) {
+      best_imp = imp;
+      best_b = b;
+    }
+  }
+  if (best_b < 0) return;
+  unsigned long long p =
+      ((unsigned long long)enc_f32(best_imp) << 32) |
+      ((unsigned long long)f << 16) | (unsigned)best_b;
+  atomicMin(&best[node], p);
+}
+
+extern "C" void launch_rf_split(const int* hist, const unsigned char* fsel,
+                                unsigned long long* best, int* cnt, int L,
+                                int C, hipStream_t stream) {
+  const int block = 256;
+  long long work = (long long)L * 12;
+  dim3 grid((unsigned)((work + block - 1) / block));
+#define RFS_CASE(CV)                                                        \
+  case CV:                                                                  \
+    hipLaunchKernelGGL((rf_split_kernel<CV>), grid, dim3(block), 0, stream, \
+                       hist, fsel, best, cnt, L);                           \
+    return;
+  switch (C) {
+    RFS_CASE(2) RFS_CASE(3) RFS_CASE(4) RFS_CASE(5) RFS_CASE(6) RFS_CASE(7)
+    RFS_CASE(8) RFS_CASE(12) RFS_CASE(16)
+    default: break;
+  }
+#undef RFS_CASE
 }
 
 // ---------------------------------------------------------------------------

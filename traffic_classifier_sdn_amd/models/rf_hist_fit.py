@@ -64,6 +64,8 @@ def _level_split_search(B, Y, nid, L, C, mf, g, device):
     f_all = np.empty(L, dtype=np.int64)
     b_all = np.empty(L, dtype=np.int64)
     featsel = torch.argsort(torch.rand(L, F, generator=g), dim=1)[:, :mf]
+    if B.is_cuda:
+        return _level_split_search_gpu(B, Y, nid, L, C, featsel, device)
     for lo in range(0, L, _CHUNK_NODES):
         hi = min(L, lo + _CHUNK_NODES)
         Lc = hi - lo
@@ -100,6 +102,48 @@ def _level_split_search(B, Y, nid, L, C, mf, g, device):
         imp_all[lo:hi] = best_imp.cpu().numpy()
         f_all[lo:hi] = (best_fb // 256).cpu().numpy()
         b_all[lo:hi] = (best_fb % 256).cpu().numpy()
+    return cnt_all, imp_all, f_all, b_all
+
+
+def _dec_key(u):
+    """Inverse of the kernel's order-preserving f32 encode."""
+    hi = (u >> 32).astype(np.uint32)
+    pos = hi >= 0x80000000
+    bits = np.where(pos, hi & 0x7FFFFFFF, ~hi)
+    return bits.astype(np.uint32).view(np.float32)
+
+
+def _level_split_search_gpu(B, Y, nid, L, C, featsel, device):
+    """Fused-kernel split search (csrc rf_split_kernel): one thread per
+    (node, feature), packed u64 atomicMin per node — replaces the torch
+    cumsum chain whose GB-scale intermediates dominated the level cost."""
+    from ..ops import gpu as og
+
+    F = B.shape[1]
+    fsel_full = torch.zeros(L, F, dtype=torch.uint8)
+    fsel_full.scatter_(1, featsel, 1)
+    fsel_dev = fsel_full.to(device)
+    cnt_all = np.empty((L, C), dtype=np.float64)
+    imp_all = np.empty(L, dtype=np.float64)
+    f_all = np.empty(L, dtype=np.int64)
+    b_all = np.empty(L, dtype=np.int64)
+    for lo in range(0, L, _CHUNK_NODES):
+        hi = min(L, lo + _CHUNK_NODES)
+        Lc = hi - lo
+        if L <= _CHUNK_NODES:
+            nidw = nid
+        else:
+            nidw = torch.where((nid >= lo) & (nid < hi), nid - lo, torch.full_like(nid, -1))
+        hist = ops.rf_hist(B, Y, nidw, Lc, C)  # int32 [Lc, F, 256, C]
+        best, cnt = og._ext.rf_split(hist.contiguous(), fsel_dev[lo:hi].contiguous())
+        del hist
+        u = best.cpu().numpy().view(np.uint64)
+        valid = u != np.uint64(0xFFFFFFFFFFFFFFFF)
+        imp = np.where(valid, _dec_key(u).astype(np.float64), np.inf)
+        imp_all[lo:hi] = imp
+        f_all[lo:hi] = np.where(valid, (u >> np.uint64(16)) & np.uint64(0xFFFF), 0).astype(np.int64)
+        b_all[lo:hi] = np.where(valid, u & np.uint64(0xFFFF), 0).astype(np.int64)
+        cnt_all[lo:hi] = cnt.cpu().numpy()
     return cnt_all, imp_all, f_all, b_all
 
 

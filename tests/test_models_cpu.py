@@ -298,3 +298,28 @@ def test_knn_proba_small_reference_sums_to_one():
     m = KNeighborsClassifier(n_neighbors=5).fit(X, y)
     p = m.predict_proba(rng.normal(size=(7, 12)))
     np.testing.assert_allclose(p.sum(axis=1), 1.0, atol=1e-12)
+
+
+def test_rf_hist_builder_deterministic():
+    """Batched level-synchronous builder: identical forest for identical
+    seed, run to run (bootstrap per tree id; one shared featsel stream)."""
+    import torch
+
+    from traffic_classifier_sdn_amd.models.rf_hist_fit import build_forest_hist
+
+    rng = np.random.default_rng(5)
+    X = torch.from_numpy(rng.normal(size=(3000, 12)).astype(np.float32))
+    y = torch.from_numpy(rng.integers(0, 6, size=3000))
+    f1 = build_forest_hist(X, y, 6, 4, seed=7)
+    f2 = build_forest_hist(X, y, 6, 4, seed=7)
+    assert len(f1) == len(f2) == 4
+    for t1, t2 in zip(f1, f2):
+        for k in ("left", "right", "feature", "threshold", "values"):
+            np.testing.assert_array_equal(t1[k], t2[k])
+    # different seed -> different forest
+    f3 = build_forest_hist(X, y, 6, 4, seed=8)
+    assert any(
+        t1["feature"].shape != t3["feature"].shape
+        or not np.array_equal(t1["feature"], t3["feature"])
+        for t1, t3 in zip(f1, f3)
+    )

@@ -59,6 +59,8 @@ void launch_smo_row(const float*, const unsigned long long*, const double*,
                     float*, float, long long, hipStream_t);
 void launch_rf_split(const int*, const unsigned char*, unsigned long long*,
                      int*, int, int, hipStream_t);
+void launch_rf_partition(const unsigned char*, int*, const int*, const int*,
+                         const int*, long long, hipStream_t);
 void launch_smo_select2(const float*, const double*, const double*,
                         const float*, unsigned long long*, const double*,
                         double, long long, hipStream_t);
@@ -339,6 +341,20 @@ static void smo_update_dev(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
                         cur_stream());
 }
 
+static void rf_partition(torch::Tensor B, torch::Tensor nid,
+                         torch::Tensor lmap, torch::Tensor feat,
+                         torch::Tensor binthr) {
+  CHECK_IN(B, torch::kUInt8);
+  CHECK_IN(nid, torch::kInt32);
+  CHECK_IN(lmap, torch::kInt32);
+  CHECK_IN(feat, torch::kInt32);
+  CHECK_IN(binthr, torch::kInt32);
+  TORCH_CHECK(B.size(1) == 12, "B must be (n,12)");
+  launch_rf_partition(B.data_ptr<unsigned char>(), nid.data_ptr<int>(),
+                      lmap.data_ptr<int>(), feat.data_ptr<int>(),
+                      binthr.data_ptr<int>(), B.size(0), cur_stream());
+}
+
 static std::vector<torch::Tensor> rf_split(torch::Tensor hist,
                                            torch::Tensor fsel) {
   CHECK_IN(hist, torch::kInt32);
@@ -434,6 +450,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("smo_solve", &smo_solve, "device-side SMO pair solve (fused iteration)");
   m.def("smo_update_dev", &smo_update_dev, "gradient update from device sol buffer");
   m.def("rf_split", &rf_split, "fused gini split search over a level histogram");
+  m.def("rf_partition", &rf_partition, "fused frontier row partition");
   m.def("smo_row", &smo_row, "K(x_i, .) kernel row for WSS-2");
   m.def("smo_select2", &smo_select2, "WSS-2 second-order j selection");
   m.def("smo_solve2", &smo_solve2, "device-side WSS-2 pair solve");

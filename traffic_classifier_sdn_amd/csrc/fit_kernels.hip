@@ -363,6 +363,43 @@ __global__ void rf_split_kernel(const int* __restrict__ hist,
   atomicMin(&best[node], p);
 }
 
+// Fused frontier partition: one pass re-assigns every row to its child
+// node (or -1 once it reaches a leaf), replacing the ~8 separate torch
+// index/gather/where passes the builder ran per level.
+//   lmap[id]   new local id of the LEFT child (-1: node became a leaf)
+//   feat[id]   split feature of node id (valid where lmap >= 0)
+//   binthr[id] split bin (go left when bin(x) <= binthr)
+__global__ void rf_partition_kernel(const unsigned char* __restrict__ B,
+                                    int* __restrict__ nid,
+                                    const int* __restrict__ lmap,
+                                    const int* __restrict__ feat,
+                                    const int* __restrict__ binthr,
+                                    long long n) {
+  constexpr int F = 12;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       row < n; row += stride) {
+    int id = nid[row];
+    if (id < 0) continue;
+    int nl = lmap[id];
+    if (nl < 0) {
+      nid[row] = -1;
+      continue;
+    }
+    int b = B[row * F + feat[id]];
+    nid[row] = nl + (b <= binthr[id] ? 0 : 1);
+  }
+}
+
+extern "C" void launch_rf_partition(const unsigned char* B, int* nid,
+                                    const int* lmap, const int* feat,
+                                    const int* binthr, long long n,
+                                    hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(rf_partition_kernel, dim3(ts_grid(n, block)), dim3(block),
+                     0, stream, B, nid, lmap, feat, binthr, n);
+}
+
 extern "C" void launch_rf_split(const int* hist, const unsigned char* fsel,
                                 unsigned long long* best, int* cnt, int L,
                                 int C, hipStream_t stream) {

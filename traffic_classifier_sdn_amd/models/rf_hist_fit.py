@@ -262,16 +262,29 @@ def build_forest_hist(
         # row partition: combined local id -> new local id of the LEFT child
         lmap = np.full(L, -1, dtype=np.int64)
         lmap[can] = 2 * np.arange(n_split, dtype=np.int64)
-        lmap_t = torch.from_numpy(lmap).to(device)
-        feat_t = torch.from_numpy(np.maximum(feat, 0)).to(device)
-        bin_t = torch.from_numpy(best_b).to(device)
-        live = nid >= 0
-        nid_l = nid.long().clamp(min=0)
-        splitting = live & (lmap_t[nid_l] >= 0)
-        vals = Ball.gather(1, feat_t[nid_l].unsqueeze(1)).squeeze(1).long()
-        go_left = vals <= bin_t[nid_l]
-        child = lmap_t[nid_l] + torch.where(go_left, 0, 1)
-        nid = torch.where(splitting, child.to(torch.int32), torch.full_like(nid, -1))
+        if Ball.is_cuda:
+            # one fused pass (csrc rf_partition_kernel) instead of ~8 torch
+            # index/gather/where kernels over the T*n rows
+            from ..ops import gpu as og
+
+            og._ext.rf_partition(
+                Ball,
+                nid,
+                torch.from_numpy(lmap.astype(np.int32)).to(device),
+                torch.from_numpy(np.maximum(feat, 0).astype(np.int32)).to(device),
+                torch.from_numpy(best_b.astype(np.int32)).to(device),
+            )
+        else:
+            lmap_t = torch.from_numpy(lmap).to(device)
+            feat_t = torch.from_numpy(np.maximum(feat, 0)).to(device)
+            bin_t = torch.from_numpy(best_b).to(device)
+            live = nid >= 0
+            nid_l = nid.long().clamp(min=0)
+            splitting = live & (lmap_t[nid_l] >= 0)
+            vals = Ball.gather(1, feat_t[nid_l].unsqueeze(1)).squeeze(1).long()
+            go_left = vals <= bin_t[nid_l]
+            child = lmap_t[nid_l] + torch.where(go_left, 0, 1)
+            nid = torch.where(splitting, child.to(torch.int32), torch.full_like(nid, -1))
         # next frontier, combined-local order [l0, r0, l1, r1, ...]
         nxt_gid = np.empty(2 * n_split, dtype=np.int64)
         nxt_gid[0::2] = kids

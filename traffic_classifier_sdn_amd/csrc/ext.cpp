@@ -49,7 +49,8 @@ void launch_flow_features(const double*, const double*, const double*, float*,
 void launch_smo_select(const float*, const double*, const double*, double,
                        long long, unsigned long long*, hipStream_t);
 void launch_rf_hist(const unsigned char*, const unsigned char*, const int*,
-                    unsigned*, long long, int, hipStream_t);
+                    const unsigned char*, unsigned*, long long, int,
+                    hipStream_t);
 void launch_smo_solve(const float*, const float*, double*, const double*,
                       unsigned long long*, float*, double*, double, double,
                       float, hipStream_t);
@@ -297,7 +298,8 @@ static void smo_select(torch::Tensor y, torch::Tensor alpha,
 }
 
 static void rf_hist(torch::Tensor bins, torch::Tensor y, torch::Tensor nid,
-                    torch::Tensor hist) {
+                    torch::Tensor hist,
+                    c10::optional<torch::Tensor> fsel = c10::nullopt) {
   CHECK_IN(bins, torch::kUInt8);
   CHECK_IN(y, torch::kUInt8);
   CHECK_IN(nid, torch::kInt32);
@@ -305,8 +307,15 @@ static void rf_hist(torch::Tensor bins, torch::Tensor y, torch::Tensor nid,
   TORCH_CHECK(bins.size(1) == 12, "bins must be (n,12)");
   TORCH_CHECK(hist.dim() == 4 && hist.size(1) == 12 && hist.size(2) == 256,
               "hist must be (nodes,12,256,C)");
+  const unsigned char* fs = nullptr;
+  if (fsel.has_value()) {
+    CHECK_IN(fsel.value(), torch::kUInt8);
+    TORCH_CHECK(fsel.value().numel() == hist.size(0) * 12,
+                "fsel must be (nodes,12)");
+    fs = fsel.value().data_ptr<unsigned char>();
+  }
   launch_rf_hist(bins.data_ptr<unsigned char>(), y.data_ptr<unsigned char>(),
-                 nid.data_ptr<int>(),
+                 nid.data_ptr<int>(), fs,
                  reinterpret_cast<unsigned*>(hist.data_ptr<int>()),
                  bins.size(0), hist.size(3), cur_stream());
 }
@@ -446,7 +455,9 @@ static void smo_update(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("smo_select", &smo_select, "WSS-1 pair candidate selection");
   m.def("smo_update", &smo_update, "fused RBF-row gradient update");
-  m.def("rf_hist", &rf_hist, "per-node per-feature class histograms (tree build)");
+  m.def("rf_hist", &rf_hist, "per-node per-feature class histograms (tree build)",
+        py::arg("bins"), py::arg("y"), py::arg("nid"), py::arg("hist"),
+        py::arg("fsel") = c10::nullopt);
   m.def("smo_solve", &smo_solve, "device-side SMO pair solve (fused iteration)");
   m.def("smo_update_dev", &smo_update_dev, "gradient update from device sol buffer");
   m.def("rf_split", &rf_split, "fused gini split search over a level histogram");

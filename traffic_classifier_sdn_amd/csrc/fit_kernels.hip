@@ -312,6 +312,13 @@ __global__ void rf_split_kernel(const int* __restrict__ hist,
   if (idx >= (long long)L * F) return;
   int node = (int)(idx / F);
   int f = (int)(idx % F);
+  // the scatter may have filled only the node's SELECTED feature planes
+  // (launch_rf_hist with fsel), so unselected threads exit before touching
+  // totals, and the node's class counts are written by its FIRST selected
+  // feature
+  if (!fsel[node * F + f]) return;
+  int first_sel = 0;
+  while (first_sel < F && !fsel[node * F + first_sel]) ++first_sel;
   const int* h = hist + ((long long)node * F + f) * 256 * C;
   float total[C];
 #pragma unroll
@@ -322,11 +329,10 @@ __global__ void rf_split_kernel(const int* __restrict__ hist,
   float n_node = 0.f;
 #pragma unroll
   for (int c = 0; c < C; ++c) n_node += total[c];
-  if (f == 0) {
+  if (f == first_sel) {
 #pragma unroll
     for (int c = 0; c < C; ++c) cnt[node * C + c] = (int)total[c];
   }
-  if (!fsel[node * F + f]) return;
   float cum[C];
 #pragma unroll
   for (int c = 0; c < C; ++c) cum[c] = 0.f;
@@ -825,9 +831,16 @@ extern "C" void launch_smo_update(const float* X, const float* y, double* grad,
 // across the whole row set and measured negligible next to the feature
 // gather.  bins are u8 (256-bin quantile grid), classes <= 16.
 // ---------------------------------------------------------------------------
+// FSEL: when a per-node mtry mask is supplied (u8 [nodes,12], round 2) the
+// scatter only touches the node's CANDIDATE features — with sklearn's
+// max_features=3 that is 3 atomics per row instead of 12, and the split
+// search never reads the unselected planes.  fsel == nullptr keeps the
+// full-12 behaviour (the parity tests' contract).
+template <bool FSEL>
 __global__ void rf_hist_kernel(const unsigned char* __restrict__ bins,  // [n,12]
                                const unsigned char* __restrict__ y,    // [n]
                                const int* __restrict__ nid,            // [n]
+                               const unsigned char* __restrict__ fsel, // [nodes,12] or null
                                unsigned* __restrict__ hist,  // [nodes,12,256,C]
                                long long n, int C) {
   constexpr int F = 12;
@@ -838,20 +851,25 @@ __global__ void rf_hist_kernel(const unsigned char* __restrict__ bins,  // [n,12
     if (node < 0) continue;
     int cls = y[t];
     const unsigned char* b = bins + t * F;
-    unsigned* base = hist + ((long long)node * F * 256 + cls) * 1;  // indexed below
 #pragma unroll
     for (int f = 0; f < F; ++f) {
+      if (FSEL && !fsel[node * F + f]) continue;
       long long cell = (((long long)node * F + f) * 256 + b[f]) * C + cls;
       atomicAdd(&hist[cell], 1u);
     }
-    (void)base;
   }
 }
 
 extern "C" void launch_rf_hist(const unsigned char* bins, const unsigned char* y,
-                               const int* nid, unsigned* hist, long long n,
-                               int C, hipStream_t stream) {
+                               const int* nid, const unsigned char* fsel,
+                               unsigned* hist, long long n, int C,
+                               hipStream_t stream) {
   const int block = 256;
-  hipLaunchKernelGGL(rf_hist_kernel, dim3(ts_grid(n, block)), dim3(block), 0,
-                     stream, bins, y, nid, hist, n, C);
+  if (fsel)
+    hipLaunchKernelGGL((rf_hist_kernel<true>), dim3(ts_grid(n, block)),
+                       dim3(block), 0, stream, bins, y, nid, fsel, hist, n, C);
+  else
+    hipLaunchKernelGGL((rf_hist_kernel<false>), dim3(ts_grid(n, block)),
+                       dim3(block), 0, stream, bins, y, nid, nullptr, hist, n,
+                       C);
 }

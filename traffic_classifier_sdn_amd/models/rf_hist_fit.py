@@ -32,6 +32,23 @@ import torch
 
 from .. import ops
 
+import os as _os
+import time as _time
+
+_TIMING = _os.environ.get("TCSDN_RF_TIMING", "") not in ("", "0")
+_tacc = {}
+
+
+def _tick(name, t0, device):
+    if not _TIMING:
+        return 0.0
+    if device is not None and str(device).startswith("cuda"):
+        torch.cuda.synchronize()
+    now = _time.perf_counter()
+    _tacc[name] = _tacc.get(name, 0.0) + (now - t0)
+    return now
+
+
 N_BINS = 256
 _CHUNK_NODES = 8192  # per rf_hist pass: 8192 * 73.7 KB ~ 600 MB of histogram (each extra chunk re-scans ALL rows, so size the chunk to hold a whole combined multi-tree frontier)
 
@@ -134,8 +151,10 @@ def _level_split_search_gpu(B, Y, nid, L, C, featsel, device):
             nidw = nid
         else:
             nidw = torch.where((nid >= lo) & (nid < hi), nid - lo, torch.full_like(nid, -1))
-        hist = ops.rf_hist(B, Y, nidw, Lc, C)  # int32 [Lc, F, 256, C]
-        best, cnt = og._ext.rf_split(hist.contiguous(), fsel_dev[lo:hi].contiguous())
+        fsel_c = fsel_dev[lo:hi].contiguous()
+        # mtry-masked scatter: only the 3 candidate feature planes per node
+        hist = og.rf_hist(B, Y, nidw, Lc, C, fsel=fsel_c)  # int32 [Lc,F,256,C]
+        best, cnt = og._ext.rf_split(hist.contiguous(), fsel_c)
         del hist
         u = best.cpu().numpy().view(np.uint64)
         valid = u != np.uint64(0xFFFFFFFFFFFFFFFF)
@@ -172,6 +191,7 @@ def build_forest_hist(
     child emission), so per-tree records fall out by masking at the end.
     """
     device = X.device
+    _ts = _time.perf_counter() if _TIMING else 0.0
     n, F = X.shape
     seed0 = 0 if seed is None else seed
     bins, edges = quantize(X, seed=seed0)
@@ -185,15 +205,34 @@ def build_forest_hist(
     # per-tree bootstrap resample (same per-tree generators/seeds as the
     # sequential builder so resamples are reproducible per tree id)
     if bootstrap:
-        rows = torch.cat(
-            [
-                torch.randint(
-                    0, n, (n,),
-                    generator=torch.Generator(device="cpu").manual_seed(seed0 + 1000003 * t),
-                )
-                for t in tree_list
-            ]
-        ).to(device)
+        if str(device).startswith("cuda"):
+            # generate the T*n resample indices on device (the CPU randint
+            # path measured 100+ ms of the 290 ms step); per-tree generator
+            # seeds keep resamples deterministic per tree id, so the
+            # tree-parallel distributed fit stays rank-independent
+            rows = torch.cat(
+                [
+                    torch.randint(
+                        0, n, (n,), device=device,
+                        generator=torch.Generator(device=device).manual_seed(
+                            seed0 + 1000003 * t
+                        ),
+                    )
+                    for t in tree_list
+                ]
+            )
+        else:
+            rows = torch.cat(
+                [
+                    torch.randint(
+                        0, n, (n,),
+                        generator=torch.Generator(device="cpu").manual_seed(
+                            seed0 + 1000003 * t
+                        ),
+                    )
+                    for t in tree_list
+                ]
+            ).to(device)
         Ball = bins[rows].contiguous()
         Yall = y8[rows].contiguous()
     else:
@@ -219,6 +258,7 @@ def build_forest_hist(
     lvl_left: List[np.ndarray] = []
     lvl_right: List[np.ndarray] = []
 
+    t0 = _tick("setup", _ts, device) if _TIMING else 0.0
     for depth in range(max_depth):
         L = len(f_tree)
         if L == 0:
@@ -226,6 +266,7 @@ def build_forest_hist(
         cnt, best_imp, best_f, best_b = _level_split_search(
             Ball, Yall, nid, L, n_classes, max_features, gsel, device
         )
+        t0 = _tick("split_search", t0, device)
         n_node = cnt.sum(axis=1)
         pp = cnt / np.maximum(n_node, 1.0)[:, None]
         parent_gini = 1.0 - (pp * pp).sum(axis=1)
@@ -295,8 +336,10 @@ def build_forest_hist(
         f_gid = nxt_gid
         f_tree = nxt_tree
         tot += 2 * counts_t
+        t0 = _tick("partition+bookkeeping", t0, device)
 
     # split the combined records per tree and flatten each
+    t0 = _tick("levels_tail", t0, device) if _TIMING else 0.0
     trees: List[Dict[str, np.ndarray]] = []
     for t in range(T):
         ids_l, vals_l, feat_l, thr_l, left_l, right_l = [], [], [], [], [], []
@@ -313,6 +356,13 @@ def build_forest_hist(
         trees.append(
             _flatten_tree(int(tot[t]), n_classes, ids_l, vals_l, feat_l, thr_l, left_l, right_l)
         )
+    if _TIMING:
+        _tick("flatten", t0, device)
+        import sys as _sys
+
+        print("RF_TIMING " + " ".join(f"{k}={v*1e3:.1f}ms" for k, v in _tacc.items()),
+              file=_sys.stderr, flush=True)
+        _tacc.clear()
     return trees
 
 

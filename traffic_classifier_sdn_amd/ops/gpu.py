@@ -195,11 +195,17 @@ def svc_predict(
     )
 
 
-def rf_hist(bins: torch.Tensor, y: torch.Tensor, nid: torch.Tensor, n_nodes: int, n_classes: int) -> torch.Tensor:
+def rf_hist(bins: torch.Tensor, y: torch.Tensor, nid: torch.Tensor, n_nodes: int,
+            n_classes: int, fsel: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Per-node per-feature class histograms.  With ``fsel`` (u8
+    [n_nodes,12] mtry mask) only the selected feature planes are scattered
+    — 3 atomics per row instead of 12 at sklearn's max_features=3; the
+    unselected planes stay zero and must not be read."""
     if bins.shape[1] != 12:  # HIP kernel is specialised for the 12-feature schema
         return _cpu.rf_hist(bins, y, nid, n_nodes, n_classes)
     hist = torch.zeros(n_nodes, 12, 256, n_classes, dtype=torch.int32, device=bins.device)
-    _ext.rf_hist(bins.contiguous(), y.contiguous(), nid.contiguous(), hist)
+    _ext.rf_hist(bins.contiguous(), y.contiguous(), nid.contiguous(), hist,
+                 None if fsel is None else fsel.contiguous())
     return hist
 
 

@@ -62,6 +62,12 @@ void launch_rf_split(const int*, const unsigned char*, unsigned long long*,
                      int*, int, int, hipStream_t);
 void launch_rf_partition(const unsigned char*, int*, const int*, const int*,
                          const int*, long long, hipStream_t);
+void launch_rf_hist_compact(const unsigned char*, const unsigned char*,
+                            const int*, const unsigned char*, unsigned*,
+                            long long, int, int, hipStream_t);
+void launch_rf_split_compact(const int*, const unsigned char*,
+                             unsigned long long*, int*, int, int, int,
+                             hipStream_t);
 void launch_smo_select2(const float*, const double*, const double*,
                         const float*, unsigned long long*, const double*,
                         double, long long, hipStream_t);
@@ -350,6 +356,39 @@ static void smo_update_dev(torch::Tensor X, torch::Tensor y, torch::Tensor grad,
                         cur_stream());
 }
 
+static std::vector<torch::Tensor> rf_level_compact(torch::Tensor bins,
+                                                   torch::Tensor y,
+                                                   torch::Tensor nid,
+                                                   torch::Tensor frank,
+                                                   torch::Tensor fidx, int64_t L,
+                                                   int64_t C) {
+  // fused level pass: compact mtry-plane histogram scatter + split search
+  CHECK_IN(bins, torch::kUInt8);
+  CHECK_IN(y, torch::kUInt8);
+  CHECK_IN(nid, torch::kInt32);
+  CHECK_IN(frank, torch::kUInt8);
+  CHECK_IN(fidx, torch::kUInt8);
+  TORCH_CHECK(bins.size(1) == 12, "bins must be (n,12)");
+  TORCH_CHECK(frank.numel() == L * 12, "frank must be (L,12)");
+  int mf = (int)(fidx.numel() / L);
+  TORCH_CHECK(mf >= 1 && (int64_t)mf * L == fidx.numel(), "fidx must be (L,mf)");
+  auto hist = torch::zeros({L, mf, 256, C},
+                           bins.options().dtype(torch::kInt32));
+  launch_rf_hist_compact(bins.data_ptr<unsigned char>(),
+                         y.data_ptr<unsigned char>(), nid.data_ptr<int>(),
+                         frank.data_ptr<unsigned char>(),
+                         reinterpret_cast<unsigned*>(hist.data_ptr<int>()),
+                         bins.size(0), (int)C, mf, cur_stream());
+  auto best = torch::full({L}, -1, bins.options().dtype(torch::kInt64));
+  auto cnt = torch::zeros({L, C}, bins.options().dtype(torch::kInt32));
+  launch_rf_split_compact(hist.data_ptr<int>(),
+                          fidx.data_ptr<unsigned char>(),
+                          reinterpret_cast<unsigned long long*>(best.data_ptr<int64_t>()),
+                          cnt.data_ptr<int>(), (int)L, (int)C, mf,
+                          cur_stream());
+  return {best, cnt};
+}
+
 static void rf_partition(torch::Tensor B, torch::Tensor nid,
                          torch::Tensor lmap, torch::Tensor feat,
                          torch::Tensor binthr) {
@@ -462,6 +501,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("smo_update_dev", &smo_update_dev, "gradient update from device sol buffer");
   m.def("rf_split", &rf_split, "fused gini split search over a level histogram");
   m.def("rf_partition", &rf_partition, "fused frontier row partition");
+  m.def("rf_level_compact", &rf_level_compact,
+        "fused level pass: mtry-compact hist scatter + split search");
   m.def("smo_row", &smo_row, "K(x_i, .) kernel row for WSS-2");
   m.def("smo_select2", &smo_select2, "WSS-2 second-order j selection");
   m.def("smo_solve2", &smo_solve2, "device-side WSS-2 pair solve");

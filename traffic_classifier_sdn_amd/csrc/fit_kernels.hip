@@ -860,6 +860,130 @@ __global__ void rf_hist_kernel(const unsigned char* __restrict__ bins,  // [n,12
   }
 }
 
+// COMPACT variant (round 2): the histogram holds only each node's mf mtry
+// candidate planes — [nodes, mf, 256, C] — which keeps the whole level's
+// atomic working set inside the 256 MB Infinity Cache (the full 12-plane
+// buffer reached 600 MB at 8192-node chunks and the scatter fell to
+// HBM-latency atomics: measured 98 ms of the 112 ms level loop).
+// frank[node*12+f] = slot index of f among the node's candidates (0xff:
+// not a candidate).
+__global__ void rf_hist_compact_kernel(
+    const unsigned char* __restrict__ bins, const unsigned char* __restrict__ y,
+    const int* __restrict__ nid, const unsigned char* __restrict__ frank,
+    unsigned* __restrict__ hist, long long n, int C, int mf) {
+  constexpr int F = 12;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < n;
+       t += stride) {
+    int node = nid[t];
+    if (node < 0) continue;
+    int cls = y[t];
+    const unsigned char* b = bins + t * F;
+    const unsigned char* fr = frank + node * F;
+#pragma unroll
+    for (int f = 0; f < F; ++f) {
+      int slot = fr[f];
+      if (slot == 0xff) continue;
+      long long cell = (((long long)node * mf + slot) * 256 + b[f]) * C + cls;
+      atomicAdd(&hist[cell], 1u);
+    }
+  }
+}
+
+extern "C" void launch_rf_hist_compact(const unsigned char* bins,
+                                       const unsigned char* y, const int* nid,
+                                       const unsigned char* frank,
+                                       unsigned* hist, long long n, int C,
+                                       int mf, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(rf_hist_compact_kernel, dim3(ts_grid(n, block)),
+                     dim3(block), 0, stream, bins, y, nid, frank, hist, n, C,
+                     mf);
+}
+
+// split search over the COMPACT histogram: one thread per (node, slot);
+// fidx[node*mf+slot] = the slot's real feature id (packed into the result
+// so the host decode is unchanged).
+template <int C>
+__global__ void rf_split_compact_kernel(const int* __restrict__ hist,
+                                        const unsigned char* __restrict__ fidx,
+                                        unsigned long long* __restrict__ best,
+                                        int* __restrict__ cnt, int L, int mf) {
+  long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long long)L * mf) return;
+  int node = (int)(idx / mf);
+  int slot = (int)(idx % mf);
+  const int* h = hist + ((long long)node * mf + slot) * 256 * C;
+  float total[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) total[c] = 0.f;
+  for (int b = 0; b < 256; ++b)
+#pragma unroll
+    for (int c = 0; c < C; ++c) total[c] += (float)h[b * C + c];
+  float n_node = 0.f;
+#pragma unroll
+  for (int c = 0; c < C; ++c) n_node += total[c];
+  if (slot == 0) {
+#pragma unroll
+    for (int c = 0; c < C; ++c) cnt[node * C + c] = (int)total[c];
+  }
+  float cum[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) cum[c] = 0.f;
+  float best_imp = FLT_MAX;
+  int best_b = -1;
+  const float inv_n = 1.f / fmaxf(n_node, 1.f);
+  for (int b = 0; b < 255; ++b) {
+    float nl = 0.f;
+#pragma unroll
+    for (int c = 0; c < C; ++c) {
+      cum[c] += (float)h[b * C + c];
+      nl += cum[c];
+    }
+    float nr = n_node - nl;
+    if (nl < 1.f || nr < 1.f) continue;
+    float sl = 0.f, sr = 0.f;
+#pragma unroll
+    for (int c = 0; c < C; ++c) {
+      float pl = cum[c] / nl;
+      float pr = (total[c] - cum[c]) / nr;
+      sl = fmaf(pl, pl, sl);
+      sr = fmaf(pr, pr, sr);
+    }
+    float imp = (nl * (1.f - sl) + nr * (1.f - sr)) * inv_n;
+    if (imp < best_imp) {
+      best_imp = imp;
+      best_b = b;
+    }
+  }
+  if (best_b < 0) return;
+  unsigned long long p =
+      ((unsigned long long)enc_f32(best_imp) << 32) |
+      ((unsigned long long)fidx[node * mf + slot] << 16) | (unsigned)best_b;
+  atomicMin(&best[node], p);
+}
+
+extern "C" void launch_rf_split_compact(const int* hist,
+                                        const unsigned char* fidx,
+                                        unsigned long long* best, int* cnt,
+                                        int L, int C, int mf,
+                                        hipStream_t stream) {
+  const int block = 256;
+  long long work = (long long)L * mf;
+  dim3 grid((unsigned)((work + block - 1) / block));
+#define RFSC_CASE(CV)                                                       \
+  case CV:                                                                  \
+    hipLaunchKernelGGL((rf_split_compact_kernel<CV>), grid, dim3(block), 0, \
+                       stream, hist, fidx, best, cnt, L, mf);               \
+    return;
+  switch (C) {
+    RFSC_CASE(2) RFSC_CASE(3) RFSC_CASE(4) RFSC_CASE(5) RFSC_CASE(6)
+    RFSC_CASE(7) RFSC_CASE(8) RFSC_CASE(12) RFSC_CASE(16)
+    default: break;
+  }
+#undef RFSC_CASE
+}
+
 extern "C" void launch_rf_hist(const unsigned char* bins, const unsigned char* y,
                                const int* nid, const unsigned char* fsel,
                                unsigned* hist, long long n, int C,

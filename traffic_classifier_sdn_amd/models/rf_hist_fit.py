@@ -80,7 +80,10 @@ def _level_split_search(B, Y, nid, L, C, mf, g, device):
     imp_all = np.empty(L, dtype=np.float64)
     f_all = np.empty(L, dtype=np.int64)
     b_all = np.empty(L, dtype=np.int64)
+    tf = _time.perf_counter() if _TIMING else 0.0
     featsel = torch.argsort(torch.rand(L, F, generator=g), dim=1)[:, :mf]
+    if _TIMING:
+        _tick("ls_featsel", tf, None)
     if B.is_cuda:
         return _level_split_search_gpu(B, Y, nid, L, C, featsel, device)
     for lo in range(0, L, _CHUNK_NODES):
@@ -130,16 +133,26 @@ def _dec_key(u):
     return bits.astype(np.uint32).view(np.float32)
 
 
-def _level_split_search_gpu(B, Y, nid, L, C, featsel, device):
+def _level_split_search_gpu(B, Y, nid, L, C, featsel, device):  # noqa: C901
     """Fused-kernel split search (csrc rf_split_kernel): one thread per
     (node, feature), packed u64 atomicMin per node — replaces the torch
     cumsum chain whose GB-scale intermediates dominated the level cost."""
     from ..ops import gpu as og
 
+    t0 = _time.perf_counter() if _TIMING else 0.0
     F = B.shape[1]
-    fsel_full = torch.zeros(L, F, dtype=torch.uint8)
-    fsel_full.scatter_(1, featsel, 1)
-    fsel_dev = fsel_full.to(device)
+    mf = featsel.shape[1]
+    # frank[node, f] = slot of f among the node's mtry candidates (0xff:
+    # not a candidate); fidx[node, slot] = real feature id of the slot —
+    # together they drive the COMPACT [L, mf, 256, C] histogram whose
+    # atomic working set stays Infinity-Cache-resident (the full 12-plane
+    # buffer reached 600 MB and fell to HBM-latency atomics)
+    frank = torch.full((L, F), 0xFF, dtype=torch.uint8)
+    slots = torch.arange(mf, dtype=torch.uint8).expand(L, mf)
+    frank.scatter_(1, featsel, slots)
+    frank_dev = frank.to(device)
+    fidx_dev = featsel.to(torch.uint8).contiguous().to(device)
+    t0 = _tick("ls_fsel", t0, device)
     cnt_all = np.empty((L, C), dtype=np.float64)
     imp_all = np.empty(L, dtype=np.float64)
     f_all = np.empty(L, dtype=np.int64)
@@ -151,11 +164,11 @@ def _level_split_search_gpu(B, Y, nid, L, C, featsel, device):
             nidw = nid
         else:
             nidw = torch.where((nid >= lo) & (nid < hi), nid - lo, torch.full_like(nid, -1))
-        fsel_c = fsel_dev[lo:hi].contiguous()
-        # mtry-masked scatter: only the 3 candidate feature planes per node
-        hist = og.rf_hist(B, Y, nidw, Lc, C, fsel=fsel_c)  # int32 [Lc,F,256,C]
-        best, cnt = og._ext.rf_split(hist.contiguous(), fsel_c)
-        del hist
+        best, cnt = og._ext.rf_level_compact(
+            B, Y, nidw, frank_dev[lo:hi].contiguous(),
+            fidx_dev[lo:hi].contiguous(), Lc, C,
+        )
+        t0 = _tick("ls_hist", t0, device)
         u = best.cpu().numpy().view(np.uint64)
         valid = u != np.uint64(0xFFFFFFFFFFFFFFFF)
         imp = np.where(valid, _dec_key(u).astype(np.float64), np.inf)
@@ -163,6 +176,7 @@ def _level_split_search_gpu(B, Y, nid, L, C, featsel, device):
         f_all[lo:hi] = np.where(valid, (u >> np.uint64(16)) & np.uint64(0xFFFF), 0).astype(np.int64)
         b_all[lo:hi] = np.where(valid, u & np.uint64(0xFFFF), 0).astype(np.int64)
         cnt_all[lo:hi] = cnt.cpu().numpy()
+        t0 = _tick("ls_decode", t0, device)
     return cnt_all, imp_all, f_all, b_all
 
 

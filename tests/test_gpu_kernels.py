@@ -478,3 +478,20 @@ def test_predict_parity_wave_tiled_band(X_real):
         svc_cpu = load_model(os.path.join(REPO, "data", "ref_models", "SVC.npz"), device="cpu")
         want_s = svc_cpu.predict_index(X)
         assert (got_s == want_s).float().mean().item() > 0.999, n
+
+
+@pytest.mark.gpu
+def test_rf_hist_fsel_mask_planes():
+    """rf_hist's mtry mask: selected feature planes match the full scatter,
+    unselected planes stay zero (API kept alongside the compact level pass)."""
+    rng = np.random.default_rng(3)
+    n, nodes, C = 100_000, 17, 6
+    bins = torch.from_numpy(rng.integers(0, 256, (n, 12)).astype(np.uint8)).cuda()
+    y = torch.from_numpy(rng.integers(0, C, n).astype(np.uint8)).cuda()
+    nid = torch.from_numpy(rng.integers(-1, nodes, n).astype(np.int32)).cuda()
+    fsel = torch.from_numpy((rng.random((nodes, 12)) < 0.3).astype(np.uint8)).cuda()
+    full = og.rf_hist(bins, y, nid, nodes, C)
+    masked = og.rf_hist(bins, y, nid, nodes, C, fsel=fsel)
+    sel = fsel.bool().cpu()[:, :, None, None]
+    assert torch.equal(masked.cpu().masked_select(sel), full.cpu().masked_select(sel))
+    assert int(masked.cpu().masked_select(~sel).abs().sum()) == 0

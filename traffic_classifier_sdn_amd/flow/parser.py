@@ -12,7 +12,7 @@ traffic_classifier.py:147-165) — including ignoring non-``data`` lines.
 
 from __future__ import annotations
 
-from typing import Iterable, Iterator, Optional, Union
+from typing import Iterable, Optional, Union
 
 from .state import FlowTable
 

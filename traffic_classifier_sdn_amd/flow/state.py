@@ -27,7 +27,7 @@ from typing import Dict, Iterator, List, Optional, Tuple
 
 import numpy as np
 
-from ..utils.schema import NUM_FEATURES
+
 
 ACTIVE = "ACTIVE"
 INACTIVE = "INACTIVE"

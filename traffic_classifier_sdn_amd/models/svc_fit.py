@@ -16,7 +16,7 @@ identical control flow, so the multi-process logic is testable on gloo.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 import torch

@@ -10,11 +10,11 @@ model matrix (notebooks/1_log_Kmeans.ipynb cells 8-18).
 from __future__ import annotations
 
 import os
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 
-from .schema import CSV_HEADER_COLUMNS, FEATURE_NAMES, LABEL_COLUMN
+from .schema import CSV_HEADER_COLUMNS, FEATURE_NAMES
 
 REFERENCE_DATASET_DIR = "/root/reference/datasets"
 # compact in-repo copy (created by tools/convert_reference.py) so GPU boxes

@@ -254,3 +254,23 @@ def test_prometheus_metrics_endpoint():
     assert "tcsdn_predict_passes_total" in body
     assert "tcsdn_predict_seconds_bucket" in body
     assert "tcsdn_class_flows" in body
+
+
+def test_serve_models_tolerate_absurd_magnitudes():
+    """A hostile/buggy switch can report absurd counters; every serve-path
+    model must classify (any label) without crashing, NaN-ing or hanging."""
+    from traffic_classifier_sdn_amd.models import load_model
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    X = np.zeros((8, 12), dtype=np.float64)
+    X[0] = 1e38          # near-f32-max rates
+    X[1] = -1e38         # nonsense negative counters
+    X[2] = 1e-40         # subnormal territory
+    X[3, 5] = np.inf     # a single inf feature
+    X[4] = 0.0
+    for name in ("RandomForestClassifier", "GaussianNB", "LogisticRegression",
+                 "SVC", "KMeans_Clustering"):
+        m = load_model(os.path.join(repo, "data", "ref_models", name + ".npz"))
+        pred = m.predict_index(X)
+        assert pred.shape[0] == 8
+        assert int(pred.min()) >= 0, name  # a real class, never a sentinel

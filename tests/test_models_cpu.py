@@ -323,3 +323,22 @@ def test_rf_hist_builder_deterministic():
         or not np.array_equal(t1["feature"], t3["feature"])
         for t1, t3 in zip(f1, f3)
     )
+
+
+def test_knn_tolerates_absurd_magnitudes():
+    """Hostile counter magnitudes (inf / 1e38 features) must yield valid
+    labels, not sentinel indices, through the brute-force path."""
+    from traffic_classifier_sdn_amd.models import KNeighborsClassifier
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(500, 12)) * 100
+    y = rng.integers(0, 6, size=500)
+    m = KNeighborsClassifier(n_neighbors=5).fit(X, y)
+    Q = np.zeros((6, 12))
+    Q[0] = 1e38
+    Q[1, 3] = np.inf
+    Q[2] = -1e38
+    Q[3] = np.nan
+    pred = m.predict_index(Q)
+    assert pred.shape[0] == 6
+    assert int(pred.min()) >= 0 and int(pred.max()) < 6

@@ -35,9 +35,21 @@ class KNeighborsClassifier(Estimator):
         self.y_: Optional[torch.Tensor] = None
         self.sharded_ = False
 
+    # |feature| cap: keeps f32 squared distances finite for hostile/absurd
+    # counter magnitudes (1e18^2 is representable, inf^2 is not); real flow
+    # features are orders of magnitude below the cap, so sane data is
+    # untouched and the brute-force ordering is preserved
+    _FEATURE_CAP = 1e18
+
+    def _ingest(self, X: ArrayLike) -> torch.Tensor:
+        t = as_tensor(X, self.device, torch.float32)
+        return torch.nan_to_num(t, nan=0.0, posinf=self._FEATURE_CAP,
+                                neginf=-self._FEATURE_CAP).clamp_(
+            -self._FEATURE_CAP, self._FEATURE_CAP)
+
     def fit(self, X: ArrayLike, y: ArrayLike, sharded: bool = False):
         """Store the reference rows (this rank's shard when ``sharded``)."""
-        self.fit_X_ = as_tensor(X, self.device, torch.float32)
+        self.fit_X_ = self._ingest(X)
         if sharded and dist.is_initialized():
             classes_local = np.unique(np.asarray(y).ravel())
             all_classes = [None] * dist.world_size()
@@ -53,7 +65,7 @@ class KNeighborsClassifier(Estimator):
 
     def kneighbors(self, X: ArrayLike):
         """Local-shard (dist, idx) top-k, batched over query rows."""
-        Xt = as_tensor(X, self.device, torch.float32)
+        Xt = self._ingest(X)
         k = min(self.n_neighbors, self.fit_X_.shape[0])
         dists, idxs = [], []
         for lo in range(0, Xt.shape[0], self.batch_rows):

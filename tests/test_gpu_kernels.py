@@ -495,3 +495,25 @@ def test_rf_hist_fsel_mask_planes():
     sel = fsel.bool().cpu()[:, :, None, None]
     assert torch.equal(masked.cpu().masked_select(sel), full.cpu().masked_select(sel))
     assert int(masked.cpu().masked_select(~sel).abs().sum()) == 0
+
+
+@pytest.mark.gpu
+def test_knn_gpu_tolerates_absurd_magnitudes():
+    """The HIP brute-force kernels (scalar + MFMA + bf16) must return valid
+    labels for hostile feature magnitudes (the ingestion clamp keeps f32
+    squared distances finite)."""
+    from traffic_classifier_sdn_amd.models import KNeighborsClassifier
+    from traffic_classifier_sdn_amd.utils.datasets import synthetic_flow_rows
+
+    rng = np.random.default_rng(0)
+    for n_ref, approx in ((50_000, False), (200_000, False), (200_000, True)):
+        X = synthetic_flow_rows(n_ref, seed=1)
+        y = rng.integers(0, 6, size=n_ref)
+        m = KNeighborsClassifier(n_neighbors=5, device="cuda", approx=approx).fit(X, y)
+        Q = np.zeros((8, 12))
+        Q[0] = 1e38
+        Q[1, 3] = np.inf
+        Q[2] = -1e38
+        Q[3] = np.nan
+        pred = m.predict_index(Q).cpu()
+        assert int(pred.min()) >= 0 and int(pred.max()) < 6, (n_ref, approx)

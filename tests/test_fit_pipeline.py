@@ -238,3 +238,24 @@ def test_fit_torchrun_world2_gloo(tmp_path):
     line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
     d = json.loads(line)
     assert d["accuracy"] > 0.97
+
+
+def test_fit_plots_confusion_svg(tmp_path):
+    """fit.py --plots writes a valid confusion-matrix heatmap SVG."""
+    import subprocess
+    import sys
+    import xml.etree.ElementTree as ET
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "traffic_classifier_sdn_amd.fit",
+         "--algos", "gaussiannb", "--out", str(tmp_path / "m"),
+         "--plots", str(tmp_path / "plots"), "--json"],
+        capture_output=True, text=True, timeout=300, cwd=repo,
+    )
+    assert r.returncode == 0, r.stderr[-800:]
+    svg = tmp_path / "plots" / "confusion_gaussiannb.svg"
+    assert svg.exists()
+    root = ET.parse(svg).getroot()
+    assert root.tag.endswith("svg")
+    assert open(svg).read().count("<rect") >= 37  # 6x6 cells + background

@@ -140,6 +140,8 @@ def main(argv: Optional[Sequence[str]] = None) -> int:
     ap.add_argument("--seed", type=int, default=101)
     ap.add_argument("--sklearn-pickles", action="store_true", help="also write sklearn-1.0.1-layout pickles")
     ap.add_argument("--json", action="store_true", help="machine-readable result lines")
+    ap.add_argument("--plots", default=None, metavar="DIR",
+                    help="write per-algorithm confusion-matrix heatmaps as SVG")
     ap.add_argument("--with-synth-quake", action="store_true",
                     help="add synthetic quake rows (D-ITG Quake3 replay) for a "
                          "true 6-class fit — the reference's quake CSV is not "
@@ -188,6 +190,15 @@ def main(argv: Optional[Sequence[str]] = None) -> int:
                 except NotImplementedError as e:
                     print(f"WARNING: {algo}: {e}", file=sys.stderr)
             res["checkpoint"] = npz_path
+            if args.plots:
+                from .utils.svgplot import confusion_svg
+
+                os.makedirs(args.plots, exist_ok=True)
+                confusion_svg(
+                    os.path.join(args.plots, f"confusion_{algo}.svg"),
+                    res["confusion_matrix"], list(CLASS_NAMES),
+                    f"{algo} confusion (acc {res['accuracy']:.4f})",
+                )
             pub = res.get("published_accuracy")
             # published numbers were measured on the real 6-class rows; with
             # a synthetic 6th class the comparison is informational only

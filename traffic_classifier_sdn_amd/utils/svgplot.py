@@ -105,3 +105,54 @@ def scatter_svg(
     with open(path, "w") as f:
         f.write(doc)
     return path
+
+
+def confusion_svg(path: str, cm, labels: Sequence[str], title: str) -> str:
+    """Confusion-matrix heatmap (the notebooks' seaborn heatmap, e.g.
+    notebooks/1_log_Kmeans.ipynb cell 59) as a standalone SVG."""
+    cm = np.asarray(cm, dtype=np.float64)
+    n = cm.shape[0]
+    cell = min(64, (W - 2 * PAD) // n)
+    x0 = PAD + 40
+    y0 = PAD / 2 + 24
+    parts = [
+        f'<svg xmlns="http://www.w3.org/2000/svg" width="{W}" height="{H}" '
+        f'viewBox="0 0 {W} {H}">',
+        f'<rect width="{W}" height="{H}" fill="white"/>',
+        f'<text x="{W/2}" y="18" text-anchor="middle" font-size="15" '
+        f'font-family="sans-serif">{title}</text>',
+    ]
+    vmax = cm.max() or 1.0
+    for i in range(n):
+        for j in range(n):
+            v = cm[i, j] / vmax
+            # white -> blue ramp
+            r = int(255 * (1 - 0.75 * v))
+            g = int(255 * (1 - 0.55 * v))
+            x, y = x0 + j * cell, y0 + i * cell
+            parts.append(
+                f'<rect x="{x}" y="{y}" width="{cell}" height="{cell}" '
+                f'fill="rgb({r},{g},255)" stroke="#ddd"/>'
+            )
+            tcol = "#000" if v < 0.6 else "#fff"
+            parts.append(
+                f'<text x="{x + cell/2}" y="{y + cell/2 + 4}" text-anchor="middle" '
+                f'font-size="11" font-family="sans-serif" fill="{tcol}">{int(cm[i, j])}</text>'
+            )
+    for k, lab in enumerate(labels):
+        parts.append(
+            f'<text x="{x0 + k*cell + cell/2}" y="{y0 + n*cell + 14}" '
+            f'text-anchor="middle" font-size="10" font-family="sans-serif">{lab[:6]}</text>'
+        )
+        parts.append(
+            f'<text x="{x0 - 6}" y="{y0 + k*cell + cell/2 + 3}" text-anchor="end" '
+            f'font-size="10" font-family="sans-serif">{lab[:6]}</text>'
+        )
+    parts.append(
+        f'<text x="{x0 + n*cell/2}" y="{y0 + n*cell + 30}" text-anchor="middle" '
+        f'font-size="11" font-family="sans-serif">predicted</text>'
+    )
+    parts.append("</svg>")
+    with open(path, "w") as f:
+        f.write("\n".join(parts))
+    return path
